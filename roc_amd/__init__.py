@@ -1,0 +1,20 @@
+"""roc_amd — MI355X-native distributed full-graph GNN training framework.
+
+A from-scratch rebuild of the capabilities of jiazhihao/ROC (MLSys'20)
+designed for AMD Instinct MI355X (gfx950/CDNA4):
+
+- hand-written HIP kernels (MFMA GEMM, CSR SpMM aggregation, fused
+  degree-norm, Philox dropout, fused softmax-CE + metrics, fused Adam)
+- one process per GPU, RCCL collectives over xGMI (halo exchange by
+  all_to_all, flat-bucket gradient all-reduce)
+- HBM-resident activations (288 GB/GPU) with optional host-DRAM offload
+- edge-balanced contiguous vertex partitioning with cost-model rebalance
+"""
+__version__ = "0.1.0"
+
+from . import graph  # noqa: F401
+from .graph import CSRGraph, load_lux, synthetic_graph, synthetic_dataset  # noqa: F401
+from .parallel.partition import GraphShard, build_shard, edge_balanced_bounds  # noqa: F401
+from .optim import AdamOptimizer  # noqa: F401
+from .engine import Trainer  # noqa: F401
+from .models import build_model, GCN, GraphSAGE, GIN  # noqa: F401

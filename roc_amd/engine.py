@@ -1,0 +1,96 @@
+"""Training engine: per-rank epoch loop, gradient all-reduce, metrics.
+
+The reference's train verbs (`Model::{init,forward,backward,update,
+zero_gradients,train_mode,infer_mode}`, `gnn.h:162-203`) map to:
+  Trainer.train_epoch  = zero_gradients + forward + backward + update
+  Trainer.evaluate     = infer_mode + forward (metrics)
+Weight-gradient reduction is a single fused RCCL all-reduce over a flat
+bucket (the reference instead gathers grad replicas to ONE GPU and sums
+serially, `optimizer_kernel.cu:88-94`).
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from .ops import functional as F
+from .optim import AdamOptimizer
+from .parallel.partition import GraphShard
+
+
+class Trainer:
+    def __init__(self, model, shard: GraphShard, feats: torch.Tensor,
+                 labels: torch.Tensor, mask: torch.Tensor,
+                 optimizer: AdamOptimizer, device="cpu",
+                 compute_dtype: torch.dtype = torch.float32,
+                 grad_scale: float = 1.0, group=None, seed: int = 1):
+        self.model = model.to(device)
+        self.shard = shard.to(device)
+        self.device = torch.device(device)
+        self.dtype = compute_dtype
+        self.group = group
+        self.grad_scale = grad_scale
+        lo, hi = shard.lo, shard.hi
+        self.x = feats[lo:hi].to(device=device, dtype=compute_dtype).contiguous()
+        self.labels = labels[lo:hi].to(device=device).contiguous()
+        self.mask = mask[lo:hi].to(device=device, dtype=torch.int32).contiguous()
+        self.optimizer = optimizer
+        self.epoch = 0
+        F.set_dropout_seed(seed + shard.rank * 7919)
+        self._flat_grad = None
+
+    # -- gradient all-reduce (flat bucket; weights are small) ---------------
+    def _allreduce_grads(self):
+        if self.shard.world_size == 1 or not dist.is_initialized():
+            return
+        grads = [p.grad for p in self.optimizer.params if p.grad is not None]
+        if not grads:
+            return
+        flat = torch.cat([g.reshape(-1) for g in grads])
+        dist.all_reduce(flat, group=self.group)
+        off = 0
+        for g in grads:
+            n = g.numel()
+            g.copy_(flat[off:off + n].view_as(g))
+            off += n
+
+    def _forward_loss(self):
+        logits = self.model(self.x, self.shard, self.group)
+        return F.softmax_cross_entropy(
+            logits, self.labels, self.mask, self.grad_scale)
+
+    def train_epoch(self):
+        self.model.train()
+        self.optimizer.zero_grad()
+        loss, metrics = self._forward_loss()
+        loss.backward()
+        self._allreduce_grads()
+        self.optimizer.step()
+        self.epoch += 1
+        return metrics
+
+    @torch.no_grad()
+    def evaluate(self) -> dict:
+        self.model.eval()
+        logits = self.model(self.x, self.shard, self.group)
+        _, metrics = F.softmax_cross_entropy(
+            logits, self.labels, self.mask, self.grad_scale)
+        if self.shard.world_size > 1 and dist.is_initialized():
+            dist.all_reduce(metrics, group=self.group)
+        return F.decode_metrics(metrics)
+
+    def sync(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+    def timed_epochs(self, n: int) -> float:
+        """Run n training epochs, return wall seconds (caller barriers)."""
+        self.sync()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            self.train_epoch()
+        self.sync()
+        return time.perf_counter() - t0

@@ -1,0 +1,255 @@
+"""Graph container, dataset loaders and synthetic generators.
+
+Replicates the reference's on-disk formats (ROC `load_task.cu:25-269`,
+`gnn.cc:751-872`) with an MI355X-native in-memory layout:
+
+- CSR over *in-edges*: ``rowptr[v]..rowptr[v+1]`` indexes the sources
+  ``colidx[e]`` of v's in-neighbors (int64 rowptr, int32 colidx).
+- A transpose CSR (CSC of the same matrix) is built once at load time so
+  the aggregation backward is exact on asymmetric graphs (the reference
+  assumes a symmetric graph and reuses the forward kernel,
+  `scattergather_kernel.cu:160-170`; we keep that fast path behind the
+  ``symmetric`` flag).
+
+File formats (reference parity):
+- ``<name>.add_self_edge.lux``: u32 numNodes, u64 numEdges, then numNodes
+  u64 *inclusive-end* row offsets, then numEdges u32 source ids.
+- ``<name>.feats.csv`` (one row of floats per vertex) with auto-written
+  binary cache ``<name>.feats.bin``.
+- ``<name>.label``: one int class id per line.
+- ``<name>.mask``: one of ``Train|Val|Test|None`` per line.
+"""
+from __future__ import annotations
+
+import os
+import struct
+from dataclasses import dataclass, field
+from typing import Optional
+
+import numpy as np
+import torch
+
+MASK_NONE = 0
+MASK_TRAIN = 1
+MASK_VAL = 2
+MASK_TEST = 3
+
+_MASK_STR = {"None": MASK_NONE, "Train": MASK_TRAIN, "Val": MASK_VAL, "Test": MASK_TEST}
+
+
+@dataclass
+class CSRGraph:
+    """In-edge CSR for one whole graph (or one partition's local view)."""
+
+    num_nodes: int
+    num_edges: int
+    rowptr: torch.Tensor  # int64 [num_nodes + 1]
+    colidx: torch.Tensor  # int32 [num_edges]
+    symmetric: bool = False
+    # transpose (CSC): built lazily unless symmetric
+    t_rowptr: Optional[torch.Tensor] = None
+    t_colidx: Optional[torch.Tensor] = None
+    # cached per-node in-degree (float32, clamped to >= 1 for rsqrt)
+    _indegree: Optional[torch.Tensor] = field(default=None, repr=False)
+
+    def indegree(self) -> torch.Tensor:
+        if self._indegree is None:
+            deg = (self.rowptr[1:] - self.rowptr[:-1]).to(torch.float32)
+            self._indegree = deg.clamp_(min=1.0)
+        return self._indegree
+
+    def transpose(self) -> "CSRGraph":
+        """Return (building if needed) the transpose adjacency as CSR views."""
+        if self.symmetric:
+            return self
+        if self.t_rowptr is None:
+            self.t_rowptr, self.t_colidx = build_transpose(
+                self.num_nodes, self.rowptr, self.colidx
+            )
+        return CSRGraph(
+            num_nodes=self.num_nodes,
+            num_edges=self.num_edges,
+            rowptr=self.t_rowptr,
+            colidx=self.t_colidx,
+            symmetric=False,
+        )
+
+    def to(self, device) -> "CSRGraph":
+        g = CSRGraph(
+            num_nodes=self.num_nodes,
+            num_edges=self.num_edges,
+            rowptr=self.rowptr.to(device),
+            colidx=self.colidx.to(device),
+            symmetric=self.symmetric,
+        )
+        if self.t_rowptr is not None:
+            g.t_rowptr = self.t_rowptr.to(device)
+            g.t_colidx = self.t_colidx.to(device)
+        if self._indegree is not None:
+            g._indegree = self._indegree.to(device)
+        return g
+
+
+def build_transpose(num_nodes: int, rowptr: torch.Tensor, colidx: torch.Tensor):
+    """CSC of the in-edge CSR: t_rowptr/t_colidx list, for each source u,
+    the destinations v that have u as an in-neighbor.
+
+    Works for rectangular local views too (rowptr rows < colidx id space):
+    pass num_nodes = size of the *column* id space.
+    """
+    nr = rowptr.numel() - 1
+    rp = rowptr.numpy()
+    ci = colidx.numpy()
+    counts = np.bincount(ci, minlength=num_nodes).astype(np.int64)
+    t_rowptr = np.zeros(num_nodes + 1, dtype=np.int64)
+    np.cumsum(counts, out=t_rowptr[1:])
+    # destination of each edge
+    dst = np.repeat(np.arange(nr, dtype=np.int32), np.diff(rp))
+    order = np.argsort(ci, kind="stable")
+    t_colidx = dst[order]
+    return torch.from_numpy(t_rowptr), torch.from_numpy(np.ascontiguousarray(t_colidx))
+
+
+# ---------------------------------------------------------------------------
+# Reference on-disk formats (ROC parity)
+# ---------------------------------------------------------------------------
+
+def load_lux(path: str) -> CSRGraph:
+    """Load a `.lux` CSR graph (header + inclusive-end offsets + col ids).
+
+    Format per reference `gnn.cc:758-801` / `load_task.cu:201-269`.
+    """
+    with open(path, "rb") as f:
+        num_nodes = struct.unpack("<I", f.read(4))[0]
+        num_edges = struct.unpack("<Q", f.read(8))[0]
+        raw_rows = np.fromfile(f, dtype=np.uint64, count=num_nodes)
+        colidx = np.fromfile(f, dtype=np.uint32, count=num_edges)
+    assert raw_rows.shape[0] == num_nodes and colidx.shape[0] == num_edges, (
+        f"truncated .lux file {path}"
+    )
+    # inclusive-end offsets -> standard rowptr
+    rowptr = np.zeros(num_nodes + 1, dtype=np.int64)
+    rowptr[1:] = raw_rows.astype(np.int64)
+    assert np.all(np.diff(rowptr) >= 0), "non-monotone row offsets"
+    assert rowptr[-1] == num_edges, "row offsets do not end at numEdges"
+    return CSRGraph(
+        num_nodes=num_nodes,
+        num_edges=num_edges,
+        rowptr=torch.from_numpy(rowptr),
+        colidx=torch.from_numpy(colidx.astype(np.int32)),
+    )
+
+
+def save_lux(path: str, g: CSRGraph) -> None:
+    with open(path, "wb") as f:
+        f.write(struct.pack("<I", g.num_nodes))
+        f.write(struct.pack("<Q", g.num_edges))
+        g.rowptr[1:].numpy().astype(np.uint64).tofile(f)
+        g.colidx.numpy().astype(np.uint32).tofile(f)
+
+
+def load_features(path_prefix: str, num_nodes: int, in_dim: int) -> torch.Tensor:
+    """CSV features with a binary cache (reference `load_task.cu:39-73`)."""
+    bin_path = path_prefix + ".feats.bin"
+    csv_path = path_prefix + ".feats.csv"
+    if os.path.exists(bin_path):
+        arr = np.fromfile(bin_path, dtype=np.float32, count=num_nodes * in_dim)
+    else:
+        arr = np.loadtxt(csv_path, delimiter=",", dtype=np.float32)
+        arr = np.ascontiguousarray(arr, dtype=np.float32)
+        arr.tofile(bin_path)
+    return torch.from_numpy(arr.reshape(num_nodes, in_dim).copy())
+
+
+def load_labels(path: str, num_nodes: int) -> torch.Tensor:
+    labels = np.loadtxt(path, dtype=np.int64).reshape(-1)
+    assert labels.shape[0] == num_nodes
+    return torch.from_numpy(labels)
+
+
+def load_mask(path: str, num_nodes: int) -> torch.Tensor:
+    with open(path) as f:
+        vals = [_MASK_STR[line.strip()] for line in f if line.strip()]
+    assert len(vals) == num_nodes
+    return torch.tensor(vals, dtype=torch.int32)
+
+
+# ---------------------------------------------------------------------------
+# Synthetic graphs (no-network benchmark datasets; BASELINE.json configs)
+# ---------------------------------------------------------------------------
+
+def synthetic_graph(
+    num_nodes: int,
+    num_edges: int,
+    seed: int = 1,
+    skew: float = 1.0,
+    add_self_edges: bool = True,
+) -> CSRGraph:
+    """Random directed graph with lognormal in-degree skew, self-edges added.
+
+    Column ids are sorted within each row (better gather locality, and the
+    reference's CSR is sorted the same way after construction).
+    """
+    rng = np.random.default_rng(seed)
+    n, e = int(num_nodes), int(num_edges)
+    if add_self_edges:
+        e_rand = max(e - n, 0)
+    else:
+        e_rand = e
+    # lognormal degree profile normalized to e_rand total
+    w = rng.lognormal(mean=0.0, sigma=skew, size=n)
+    w = w / w.sum()
+    deg = rng.multinomial(e_rand, w).astype(np.int64)
+    rowptr = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(deg + (1 if add_self_edges else 0), out=rowptr[1:])
+    total = int(rowptr[-1])
+    colidx = rng.integers(0, n, size=total, dtype=np.int64)
+    if add_self_edges:
+        # overwrite one slot per row with the self edge; then sort rows
+        colidx[rowptr[:-1]] = np.arange(n, dtype=np.int64)
+    # sort columns within each row via composite key
+    row_of_edge = np.repeat(np.arange(n, dtype=np.int64), np.diff(rowptr))
+    key = row_of_edge * n + colidx
+    key.sort(kind="stable")
+    colidx = (key % n).astype(np.int32)
+    return CSRGraph(
+        num_nodes=n,
+        num_edges=total,
+        rowptr=torch.from_numpy(rowptr),
+        colidx=torch.from_numpy(colidx),
+    )
+
+
+# Named shapes from BASELINE.json (synthetic stand-ins; there is no network
+# for real datasets). Node/edge counts follow the reference workloads.
+DATASET_SHAPES = {
+    # name: (num_nodes, num_edges incl. self, in_dim, num_classes)
+    "cora": (2_708, 13_264, 1_433, 7),
+    "reddit": (232_965, 114_848_857, 602, 41),
+    "amazon": (1_569_960, 132_954_714, 200, 107),
+    "ogbn-products": (2_449_029, 126_167_053, 100, 47),
+    "papers100M": (111_059_956, 1_726_745_828, 128, 172),
+    # scaled-down stand-in for the papers100M offload config (fits CI boxes)
+    "papers-synth-small": (4_000_000, 240_000_000, 128, 172),
+}
+
+
+def synthetic_dataset(name: str, seed: int = 1, scale: float = 1.0):
+    """Graph + features + labels + masks of the named shape.
+
+    ``scale`` < 1 shrinks nodes/edges proportionally (for tests).
+    Returns (graph, features fp32 [N, in_dim], labels int64 [N], mask int32 [N]).
+    """
+    n, e, d, c = DATASET_SHAPES[name]
+    n = max(int(n * scale), 16)
+    e = max(int(e * scale), n)
+    g = synthetic_graph(n, e, seed=seed)
+    rng = np.random.default_rng(seed + 1)
+    feats = torch.from_numpy(rng.standard_normal((n, d), dtype=np.float32))
+    labels = torch.from_numpy(rng.integers(0, c, size=n).astype(np.int64))
+    # 70/15/15 split like common full-graph benchmarks
+    u = rng.random(n)
+    mask = np.full(n, MASK_TRAIN, dtype=np.int32)
+    mask[u >= 0.70] = MASK_VAL
+    mask[u >= 0.85] = MASK_TEST
+    return g, feats, labels, torch.from_numpy(mask), c

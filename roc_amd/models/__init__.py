@@ -1,0 +1,14 @@
+from .gcn import GCN
+from .sage import GraphSAGE
+from .gin import GIN
+
+
+def build_model(name: str, dims, dropout: float = 0.5, seed: int = 1, **kw):
+    name = name.lower()
+    if name == "gcn":
+        return GCN(dims, dropout=dropout, seed=seed, **kw)
+    if name in ("sage", "graphsage"):
+        return GraphSAGE(dims, dropout=dropout, seed=seed, **kw)
+    if name == "gin":
+        return GIN(dims, dropout=dropout, seed=seed, **kw)
+    raise ValueError(f"unknown model {name!r}")

@@ -1,0 +1,59 @@
+"""GCN with configurable depth and optional residual connections.
+
+Mirrors the reference driver's model (`gnn.cc:66-92`): per layer
+  dropout -> linear -> indegree_norm -> scatter_gather -> indegree_norm
+  -> relu (not on last layer) [-> residual projection + add]
+The two indegree_norms around the aggregation implement the symmetric
+D^-1/2 A D^-1/2 GCN normalization; on GPU they are fused into the SpMM
+kernel (`scatter_gather(..., normalize=True)`).
+"""
+from __future__ import annotations
+
+import torch
+
+from ..ops import functional as F
+from ..ops.reference import glorot_uniform
+from ..parallel.halo import halo_exchange
+
+
+class GCN(torch.nn.Module):
+    def __init__(self, dims, dropout: float = 0.5, seed: int = 1,
+                 residual: bool = False, fused_norm: bool = True):
+        super().__init__()
+        self.dims = list(dims)
+        self.p = float(dropout)
+        self.residual = residual
+        self.fused_norm = fused_norm
+        self.weights = torch.nn.ParameterList()
+        self.res_proj = torch.nn.ParameterList()
+        for i in range(len(dims) - 1):
+            self.weights.append(torch.nn.Parameter(
+                glorot_uniform((dims[i], dims[i + 1]), seed=seed + i)))
+            if residual and dims[i] != dims[i + 1]:
+                self.res_proj.append(torch.nn.Parameter(
+                    glorot_uniform((dims[i], dims[i + 1]), seed=seed + 1000 + i)))
+            else:
+                self.res_proj.append(torch.nn.Parameter(torch.empty(0)))
+
+    def forward(self, x, shard, group=None):
+        nlayers = len(self.weights)
+        for i, w in enumerate(self.weights):
+            h = F.dropout(x, self.p, self.training)
+            h = F.linear(h, w)
+            h = halo_exchange(h, shard, group)
+            if self.fused_norm:
+                h = F.scatter_gather(h, shard, normalize=True)
+            else:
+                h = F.indegree_norm(
+                    F.scatter_gather(F.degree_scale(h, shard.rsqrt_deg_ext), shard),
+                    shard)
+            if i < nlayers - 1:
+                h = F.relu(h)
+            if self.residual:
+                proj = self.res_proj[i]
+                if proj.numel() > 0:
+                    h = F.add(h, F.linear(x, proj))
+                elif x.shape == h.shape:
+                    h = F.add(h, x)
+            x = h
+        return x

@@ -1,0 +1,44 @@
+"""GIN (Graph Isomorphism Network), sum aggregation + 2-layer MLP.
+
+h' = MLP( (1 + eps) * h_v + sum_{u in N(v)} h_u ),  eps learnable.
+Exercises the scatter-add aggregation path (BASELINE.json config #4).
+"""
+from __future__ import annotations
+
+import torch
+
+from ..ops import functional as F
+from ..ops.reference import glorot_uniform
+from ..parallel.halo import halo_exchange
+
+
+class GIN(torch.nn.Module):
+    def __init__(self, dims, dropout: float = 0.5, seed: int = 1,
+                 mlp_hidden: int = 0):
+        super().__init__()
+        self.dims = list(dims)
+        self.p = float(dropout)
+        self.eps = torch.nn.ParameterList()
+        self.w1 = torch.nn.ParameterList()
+        self.w2 = torch.nn.ParameterList()
+        for i in range(len(dims) - 1):
+            hid = mlp_hidden or dims[i + 1]
+            self.eps.append(torch.nn.Parameter(torch.zeros(1)))
+            self.w1.append(torch.nn.Parameter(
+                glorot_uniform((dims[i], hid), seed=seed + 2 * i)))
+            self.w2.append(torch.nn.Parameter(
+                glorot_uniform((hid, dims[i + 1]), seed=seed + 2 * i + 1)))
+
+    def forward(self, x, shard, group=None):
+        nlayers = len(self.w1)
+        for i in range(nlayers):
+            h = F.dropout(x, self.p, self.training)
+            hx = halo_exchange(h, shard, group)
+            agg = F.scatter_gather(hx, shard)
+            h = agg + (1.0 + self.eps[i]).to(h.dtype) * h  # scalar-eps glue
+            h = F.linear(h, self.w1[i], activation="relu")
+            h = F.linear(h, self.w2[i])
+            if i < nlayers - 1:
+                h = F.relu(h)
+            x = h
+        return x

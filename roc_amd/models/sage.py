@@ -1,0 +1,45 @@
+"""GraphSAGE (mean aggregator, concat variant).
+
+h' = relu( W_self·h_v  +  W_neigh·mean_{u in N(v)} h_u )
+
+The concat-then-linear of the original formulation is algebraically folded
+into two GEMMs (no concat materialization), and W_neigh is applied BEFORE
+the halo exchange + aggregation so the exchanged rows are already at the
+(usually smaller) output width.
+"""
+from __future__ import annotations
+
+import torch
+
+from ..ops import functional as F
+from ..ops.reference import glorot_uniform
+from ..parallel.halo import halo_exchange
+
+
+class GraphSAGE(torch.nn.Module):
+    def __init__(self, dims, dropout: float = 0.5, seed: int = 1):
+        super().__init__()
+        self.dims = list(dims)
+        self.p = float(dropout)
+        self.w_self = torch.nn.ParameterList()
+        self.w_neigh = torch.nn.ParameterList()
+        for i in range(len(dims) - 1):
+            self.w_self.append(torch.nn.Parameter(
+                glorot_uniform((dims[i], dims[i + 1]), seed=seed + 2 * i)))
+            self.w_neigh.append(torch.nn.Parameter(
+                glorot_uniform((dims[i], dims[i + 1]), seed=seed + 2 * i + 1)))
+
+    def forward(self, x, shard, group=None):
+        nlayers = len(self.w_self)
+        for i in range(nlayers):
+            h = F.dropout(x, self.p, self.training)
+            h_self = F.linear(h, self.w_self[i])
+            hn = F.linear(h, self.w_neigh[i])
+            hn = halo_exchange(hn, shard, group)
+            hn = F.scatter_gather(hn, shard)           # sum over in-neighbors
+            hn = F.degree_scale(hn, shard.inv_deg_local)  # -> mean
+            h = F.add(h_self, hn)
+            if i < nlayers - 1:
+                h = F.relu(h)
+            x = h
+        return x

@@ -1,0 +1,405 @@
+"""Autograd operators with HIP-kernel dispatch.
+
+Every op dispatches to the hand-written CDNA4 kernels in ``roc_amd._C``
+when the tensors live on a GPU, and to the plain-PyTorch fp32 reference
+(`reference.py`) on CPU. There is NO silent eager fallback on GPU: if the
+extension is missing for a CUDA tensor, the op raises.
+
+Op semantics follow the reference operator layer;
+the architecture (halo-sharded CSR, bf16 activations + fp32 masters,
+fused kernels) is MI355X-native.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+
+try:
+    from roc_amd import _C  # built by setup.py build_ext --inplace
+except ImportError:  # pragma: no cover - exercised only when ext missing
+    _C = None
+
+
+def has_ext() -> bool:
+    return _C is not None
+
+
+def _hip(x: torch.Tensor) -> bool:
+    if x.is_cuda:
+        if _C is None:
+            raise RuntimeError(
+                "roc_amd._C extension is not built but a GPU tensor was passed. "
+                "Run `python setup.py build_ext --inplace`. Refusing a silent "
+                "eager fallback on GPU."
+            )
+        return True
+    return False
+
+
+# ---------------------------------------------------------------------------
+# ScatterGather (CSR SpMM neighbor aggregation) — the hot op
+# ---------------------------------------------------------------------------
+
+class _SpMM(torch.autograd.Function):
+    """out[v] = sum over in-neighbors of x[u] on a (possibly halo-extended,
+    possibly degree-normalized) local CSR. Backward runs the same kernel on
+    the transpose CSR (exact on asymmetric graphs)."""
+
+    @staticmethod
+    def forward(ctx, x, rowptr, colidx, t_rowptr, t_colidx, num_rows, num_ext,
+                deg_dst, deg_src, row_order, t_row_order):
+        ctx.save_for_backward(t_rowptr, t_colidx, deg_dst, deg_src, t_row_order)
+        ctx.num_ext = num_ext
+        if _hip(x):
+            out = torch.empty(num_rows, x.shape[1], dtype=x.dtype, device=x.device)
+            _C.spmm(out, x, rowptr, colidx, deg_dst, deg_src, row_order)
+        else:
+            xin = x
+            if deg_src is not None:  # deg_* are precomputed row scale factors
+                xin = xin * deg_src.unsqueeze(1).to(xin.dtype)
+            out = ref.spmm(xin, rowptr, colidx, num_rows)
+            if deg_dst is not None:
+                out = out * deg_dst.unsqueeze(1).to(out.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        t_rowptr, t_colidx, deg_dst, deg_src, t_row_order = ctx.saved_tensors
+        dy = dy.contiguous()
+        # d/dx of  diag(a) A diag(b) x  =  diag(b) A^T diag(a) dy
+        if _hip(dy):
+            dx = torch.empty(ctx.num_ext, dy.shape[1], dtype=dy.dtype, device=dy.device)
+            _C.spmm(dx, dy, t_rowptr, t_colidx, deg_src, deg_dst, t_row_order)
+        else:
+            yin = dy
+            if deg_dst is not None:
+                yin = yin * deg_dst.unsqueeze(1).to(yin.dtype)
+            dx = ref.spmm(yin, t_rowptr, t_colidx, ctx.num_ext)
+            if deg_src is not None:
+                dx = dx * deg_src.unsqueeze(1).to(dx.dtype)
+        return (dx,) + (None,) * 10
+
+
+def scatter_gather(x, shard, normalize: bool = False):
+    """Neighbor sum-aggregation over the shard's local CSR.
+
+    x: [n_ext, D] halo-extended features. Returns [n_local, D].
+    normalize=True fuses the symmetric D^-1/2 A D^-1/2 GCN normalization
+    (the reference composes indegree_norm -> scatter_gather -> indegree_norm,
+    `gnn.cc:82-84`; the fused kernel reads each feature once instead).
+    """
+    deg_dst = shard.rsqrt_deg_local if normalize else None
+    deg_src = shard.rsqrt_deg_ext if normalize else None
+    return _SpMM.apply(
+        x, shard.rowptr, shard.colidx, shard.t_rowptr, shard.t_colidx,
+        shard.n_local, shard.n_ext, deg_dst, deg_src,
+        shard.row_order, shard.t_row_order,
+    )
+
+
+# ---------------------------------------------------------------------------
+# InDegreeNorm
+# ---------------------------------------------------------------------------
+
+class _DegScale(torch.autograd.Function):
+    """out = x * scale_row (rowwise). Self-adjoint: backward applies the
+    same scaling to the gradient (reference `graphnorm_kernel.cu:126-136`)."""
+
+    @staticmethod
+    def forward(ctx, x, scale):
+        ctx.save_for_backward(scale)
+        if _hip(x):
+            out = torch.empty_like(x)
+            _C.rowscale(out, x, scale)
+        else:
+            out = x * scale.unsqueeze(1).to(x.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (scale,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _hip(dy):
+            dx = torch.empty_like(dy)
+            _C.rowscale(dx, dy, scale)
+        else:
+            dx = dy * scale.unsqueeze(1).to(dy.dtype)
+        return dx, None
+
+
+def indegree_norm(x, shard):
+    """out[v] = x[v] / sqrt(indeg(v)) on local rows (reference graphnorm op)."""
+    return _DegScale.apply(x, shard.rsqrt_deg_local)
+
+
+def degree_scale(x, scale):
+    """Generic rowwise scale (used by GraphSAGE mean aggregation: 1/deg)."""
+    return _DegScale.apply(x, scale)
+
+
+# ---------------------------------------------------------------------------
+# Linear (hand-written MFMA GEMM on GPU)
+# ---------------------------------------------------------------------------
+
+class _Linear(torch.autograd.Function):
+    """y = x @ w (+ fused ReLU). w is an fp32 master [in_dim, out_dim];
+    on GPU x is bf16 and w is cast per call (w is tiny: <1 MB).
+
+    Backward: dw = x^T dy (fp32 accumulate, split-K), dx = dy @ w^T.
+    Reference: `linear_kernel.cu:20-127` (fwd+fused relu), `:130-245` (bwd).
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, act):
+        if _hip(x):
+            wc = w.to(x.dtype)
+            y = torch.empty(x.shape[0], w.shape[1], dtype=x.dtype, device=x.device)
+            _C.gemm_rr(y, x, wc, act == "relu")
+            if act == "sigmoid":
+                _C.sigmoid_fwd(y, y)
+        else:
+            y = ref.linear(x, w)
+            if act == "relu":
+                y = ref.relu(y)
+            elif act == "sigmoid":
+                y = ref.sigmoid(y)
+        ctx.save_for_backward(x, w, y if act else None)
+        ctx.act = act
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        act = ctx.act
+        dy = dy.contiguous()
+        if _hip(dy):
+            if act == "relu":
+                dym = torch.empty_like(dy)
+                _C.relu_bwd(dym, dy, y)
+                dy = dym
+            elif act == "sigmoid":
+                dym = torch.empty_like(dy)
+                _C.sigmoid_bwd(dym, dy, y)
+                dy = dym
+            dw = torch.zeros_like(w)  # fp32 [in, out]
+            _C.gemm_atb(dw, x, dy)
+            wt = w.t().contiguous().to(dy.dtype)
+            dx = torch.empty_like(x)
+            _C.gemm_rr(dx, dy, wt, False)
+        else:
+            if act == "relu":
+                dy = ref.relu_grad(dy, y)
+            elif act == "sigmoid":
+                dy = ref.sigmoid_grad(dy, y)
+            dw = (x.to(torch.float32).t() @ dy.to(torch.float32))
+            dx = (dy @ w.t().to(dy.dtype))
+        return dx, dw, None
+
+
+def linear(x, w, activation: Optional[str] = None):
+    return _Linear.apply(x, w, activation)
+
+
+# ---------------------------------------------------------------------------
+# Standalone activations (reference activation op)
+# ---------------------------------------------------------------------------
+
+class _Act(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kind):
+        if _hip(x):
+            y = torch.empty_like(x)
+            getattr(_C, f"{kind}_fwd")(y, x)
+        else:
+            y = ref.relu(x) if kind == "relu" else ref.sigmoid(x)
+        ctx.save_for_backward(y)
+        ctx.kind = kind
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _hip(dy):
+            dx = torch.empty_like(dy)
+            getattr(_C, f"{ctx.kind}_bwd")(dx, dy, y)
+        else:
+            dx = ref.relu_grad(dy, y) if ctx.kind == "relu" else ref.sigmoid_grad(dy, y)
+        return dx, None
+
+
+def relu(x):
+    return _Act.apply(x, "relu")
+
+
+def sigmoid(x):
+    return _Act.apply(x, "sigmoid")
+
+
+# ---------------------------------------------------------------------------
+# Element (add / mul) — reference element op; autograd-composable
+# ---------------------------------------------------------------------------
+
+class _Add(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        if _hip(a):
+            out = torch.empty_like(a)
+            _C.ewise_add(out, a, b)
+            return out
+        return a + b
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy, dy
+
+
+class _Mul(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        ctx.save_for_backward(a, b)
+        if _hip(a):
+            out = torch.empty_like(a)
+            _C.ewise_mul(out, a, b)
+            return out
+        return a * b
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, b = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _hip(dy):
+            da = torch.empty_like(a)
+            db = torch.empty_like(b)
+            _C.ewise_mul(da, dy, b)
+            _C.ewise_mul(db, dy, a)
+            return da, db
+        return dy * b, dy * a
+
+
+def add(a, b):
+    return _Add.apply(a, b)
+
+
+def mul(a, b):
+    return _Mul.apply(a, b)
+
+
+# ---------------------------------------------------------------------------
+# Dropout (Philox counter-based; mask regenerated in backward — no mask
+# tensor is ever materialized on GPU)
+# ---------------------------------------------------------------------------
+
+_DROPOUT_STATE = {"seed": 1, "offset": 0}
+
+
+def set_dropout_seed(seed: int) -> None:
+    _DROPOUT_STATE["seed"] = int(seed)
+    _DROPOUT_STATE["offset"] = 0
+
+
+class _Dropout(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p):
+        seed = _DROPOUT_STATE["seed"]
+        offset = _DROPOUT_STATE["offset"]
+        _DROPOUT_STATE["offset"] += 1
+        ctx.params = (p, seed, offset)
+        if _hip(x):
+            y = torch.empty_like(x)
+            _C.dropout_fwd(y, x, p, seed, offset)
+            ctx.cpu_mask = None
+            return y
+        mask = ref.dropout_mask(x.shape, p, seed, offset, x.device)
+        ctx.save_for_backward(mask)
+        return x * mask.to(x.dtype) / (1.0 - p)
+
+    @staticmethod
+    def backward(ctx, dy):
+        p, seed, offset = ctx.params
+        dy = dy.contiguous()
+        if _hip(dy):
+            dx = torch.empty_like(dy)
+            _C.dropout_fwd(dx, dy, p, seed, offset)  # same Philox stream
+            return dx, None
+        (mask,) = ctx.saved_tensors
+        return dy * mask.to(dy.dtype) / (1.0 - p), None
+
+
+def dropout(x, p: float, training: bool):
+    """Train: mask+scale; infer: identity (reference `dropout_kernel.cu:159-180`)."""
+    if not training or p <= 0.0:
+        return x
+    return _Dropout.apply(x, p)
+
+
+# ---------------------------------------------------------------------------
+# SoftmaxCrossEntropy (fused loss + grad + metrics)
+# ---------------------------------------------------------------------------
+
+class _SoftmaxCE(torch.autograd.Function):
+    """Returns (loss, metrics[8]) where metrics =
+    [roc_loss_sum, ce_loss_sum, train_correct, train_total,
+     val_correct, val_total, test_correct, test_total] (float32, on device).
+    loss = ce_loss_sum (differentiable driver for .backward()).
+    Gradient: (softmax - onehot) * (mask == Train) * grad_scale
+    (reference `softmax_kernel.cu:19-79`).
+    """
+
+    @staticmethod
+    def forward(ctx, logits, labels, mask, grad_scale):
+        if _hip(logits):
+            dl = torch.empty_like(logits)
+            metrics = torch.zeros(8, dtype=torch.float32, device=logits.device)
+            _C.softmax_ce(dl, metrics, logits, labels, mask, grad_scale)
+            loss = metrics[1].clone()
+        else:
+            dl, md = ref.softmax_cross_entropy(logits, labels, mask, grad_scale)
+            metrics = torch.tensor(
+                [md["roc_loss"], md["ce_loss"] * max(md["train_total"], 1),
+                 md["train_correct"], md["train_total"],
+                 md["val_correct"], md["val_total"],
+                 md["test_correct"], md["test_total"]],
+                dtype=torch.float32)
+            loss = metrics[1].clone()
+        ctx.save_for_backward(dl)
+        ctx.mark_non_differentiable(metrics)
+        return loss, metrics
+
+    @staticmethod
+    def backward(ctx, dloss, _dmetrics):
+        (dl,) = ctx.saved_tensors
+        return dl * dloss, None, None, None
+
+
+def softmax_cross_entropy(logits, labels, mask, grad_scale: float = 1.0):
+    return _SoftmaxCE.apply(logits, labels, mask, grad_scale)
+
+
+def decode_metrics(metrics: torch.Tensor) -> dict:
+    m = metrics.cpu().tolist()
+    tt = max(m[3], 1.0)
+    return {
+        "roc_loss": m[0],
+        "ce_loss": m[1] / tt,
+        "train_acc": m[2] / tt,
+        "val_acc": m[4] / max(m[5], 1.0),
+        "test_acc": m[6] / max(m[7], 1.0),
+        "train_total": int(m[3]),
+        "val_total": int(m[5]),
+        "test_total": int(m[7]),
+    }
+
+
+# ---------------------------------------------------------------------------
+# Adam (not autograd — called by the optimizer)
+# ---------------------------------------------------------------------------
+
+def adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay):
+    if _hip(w):
+        _C.adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay)
+    else:
+        ref.adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay)

@@ -1,0 +1,173 @@
+"""Vertex partitioner + halo plan — the explicit scheduler replacing the
+reference's Legion mapper / implicit region movement.
+
+The reference replicates the ENTIRE input tensor into every GPU through
+pinned host memory per aggregation (`scattergather.cc:68-73`,
+`types.cu:22-32`). Here each rank owns a contiguous vertex range
+(edge-balanced, like `gnn.cc:804-870`), keeps its activations resident in
+HBM, and exchanges only boundary-vertex rows ("halo") over RCCL/xGMI
+before each aggregation. The plan below precomputes, once per graph:
+
+- local CSR with columns remapped into [0, n_local + n_halo)
+- the transpose CSR (for an exact backward on asymmetric graphs)
+- per-destination send index lists (for all_to_all_single)
+- degree-scale vectors for fused GCN normalization
+- degree-descending row orders (skew-tolerant kernel scheduling)
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from ..graph import CSRGraph, build_transpose
+
+
+@dataclass
+class GraphShard:
+    """Everything one rank needs to run all graph ops locally."""
+
+    rank: int
+    world_size: int
+    bounds: List[int]            # global partition bounds, len world_size+1
+    lo: int
+    hi: int
+    n_local: int
+    n_halo: int
+    n_ext: int                   # n_local + n_halo
+    rowptr: torch.Tensor         # int64 [n_local+1]
+    colidx: torch.Tensor         # int32 [local_edges], values < n_ext
+    t_rowptr: torch.Tensor       # int64 [n_ext+1]
+    t_colidx: torch.Tensor       # int32 [local_edges], values < n_local
+    deg_local: torch.Tensor      # fp32 global in-degree of local rows (>=1)
+    rsqrt_deg_local: torch.Tensor
+    inv_deg_local: torch.Tensor
+    rsqrt_deg_ext: torch.Tensor  # fp32 [n_ext] (local ++ halo)
+    row_order: torch.Tensor      # int32 [n_local] degree-descending
+    t_row_order: torch.Tensor    # int32 [n_ext]
+    halo_ids: torch.Tensor       # int64 [n_halo] global ids (grouped by owner)
+    recv_splits: List[int]       # rows received from each rank
+    send_idx: torch.Tensor       # int64 local indices to send (concat by dest)
+    send_splits: List[int]       # rows sent to each rank
+
+    def to(self, device) -> "GraphShard":
+        d = {}
+        for k, v in self.__dict__.items():
+            d[k] = v.to(device) if isinstance(v, torch.Tensor) else v
+        return GraphShard(**d)
+
+
+def edge_balanced_bounds(rowptr: torch.Tensor, num_parts: int) -> List[int]:
+    """Contiguous vertex ranges with ~equal in-edge counts
+    (reference `gnn.cc:806-829` greedy pack; here via searchsorted)."""
+    rp = rowptr.numpy()
+    e = int(rp[-1])
+    n = rp.shape[0] - 1
+    targets = [(e * k) // num_parts for k in range(num_parts + 1)]
+    bounds = np.searchsorted(rp, targets, side="left").tolist()
+    bounds[0], bounds[-1] = 0, n
+    # ensure monotone non-decreasing (degenerate tiny graphs)
+    for i in range(1, len(bounds)):
+        bounds[i] = max(bounds[i], bounds[i - 1])
+    return bounds
+
+
+def rebalance_bounds(rowptr: torch.Tensor, bounds: List[int],
+                     per_rank_time: List[float]) -> List[int]:
+    """Cost-model refinement (the MLSys'20 Roc idea the reference code lacks):
+    re-split so each rank's predicted edge load is proportional to its
+    measured throughput (edges_k / time_k)."""
+    rp = rowptr.numpy().astype(np.float64)
+    p = len(bounds) - 1
+    edges = [rp[bounds[k + 1]] - rp[bounds[k]] for k in range(p)]
+    thr = [edges[k] / max(per_rank_time[k], 1e-9) for k in range(p)]
+    total_thr = sum(thr)
+    e_total = rp[-1]
+    targets = np.cumsum([0.0] + [e_total * t / total_thr for t in thr])
+    nb = np.searchsorted(rp, targets, side="left").tolist()
+    nb[0], nb[-1] = 0, rowptr.numel() - 1
+    for i in range(1, len(nb)):
+        nb[i] = max(nb[i], nb[i - 1])
+    return nb
+
+
+def build_shard(g: CSRGraph, rank: int, world_size: int,
+                bounds: Optional[List[int]] = None) -> GraphShard:
+    """Build this rank's shard from the full graph (all ranks deterministic)."""
+    if bounds is None:
+        bounds = edge_balanced_bounds(g.rowptr, world_size)
+    lo, hi = bounds[rank], bounds[rank + 1]
+    n_local = hi - lo
+    rp = g.rowptr.numpy()
+    ci = g.colidx.numpy()
+    e0, e1 = int(rp[lo]), int(rp[hi])
+    local_rowptr = (rp[lo:hi + 1] - e0).astype(np.int64)
+    local_cols_global = ci[e0:e1].astype(np.int64)
+
+    # halo: remote source vertices, grouped by owning rank then sorted by id
+    is_local = (local_cols_global >= lo) & (local_cols_global < hi)
+    remote = np.unique(local_cols_global[~is_local])
+    owners = np.searchsorted(bounds, remote, side="right") - 1
+    order = np.lexsort((remote, owners))
+    halo_ids = remote[order]
+    halo_owners = owners[order]
+    recv_splits = np.bincount(halo_owners, minlength=world_size).tolist()
+    n_halo = halo_ids.shape[0]
+
+    # remap columns -> ext index space [0, n_local + n_halo)
+    colidx = np.empty(local_cols_global.shape[0], dtype=np.int32)
+    colidx[is_local] = (local_cols_global[is_local] - lo).astype(np.int32)
+    pos = np.searchsorted(halo_ids, local_cols_global[~is_local])
+    colidx[~is_local] = (n_local + pos).astype(np.int32)
+
+    n_ext = n_local + n_halo
+    rowptr_t = torch.from_numpy(local_rowptr)
+    colidx_t = torch.from_numpy(colidx)
+    t_rowptr, t_colidx = build_transpose(n_ext, rowptr_t, colidx_t)
+
+    # who needs MY rows: every rank computes every rank's halo deterministically
+    send_chunks, send_splits = [], []
+    for r in range(world_size):
+        if r == rank:
+            send_chunks.append(np.empty(0, dtype=np.int64))
+            send_splits.append(0)
+            continue
+        rlo, rhi = bounds[r], bounds[r + 1]
+        their_cols = ci[rp[rlo]:rp[rhi]]
+        theirs_from_me = np.unique(
+            their_cols[(their_cols >= lo) & (their_cols < hi)]
+        ).astype(np.int64)
+        send_chunks.append(theirs_from_me - lo)
+        send_splits.append(theirs_from_me.shape[0])
+    send_idx = torch.from_numpy(np.concatenate(send_chunks)) if world_size > 1 \
+        else torch.empty(0, dtype=torch.int64)
+
+    # degrees (GLOBAL in-degree — normalization must match the 1-GPU model)
+    deg_all = np.maximum(np.diff(rp), 1).astype(np.float32)
+    deg_local = deg_all[lo:hi]
+    deg_ext = np.concatenate([deg_local, deg_all[halo_ids]]) if n_halo else deg_local
+
+    # degree-descending row orders (long rows scheduled first in the kernel)
+    local_deg_edges = np.diff(local_rowptr)
+    row_order = np.argsort(-local_deg_edges, kind="stable").astype(np.int32)
+    t_deg = np.diff(t_rowptr.numpy())
+    t_row_order = np.argsort(-t_deg, kind="stable").astype(np.int32)
+
+    return GraphShard(
+        rank=rank, world_size=world_size, bounds=list(bounds),
+        lo=lo, hi=hi, n_local=n_local, n_halo=n_halo, n_ext=n_ext,
+        rowptr=rowptr_t, colidx=colidx_t,
+        t_rowptr=t_rowptr, t_colidx=t_colidx,
+        deg_local=torch.from_numpy(deg_local.copy()),
+        rsqrt_deg_local=torch.from_numpy(1.0 / np.sqrt(deg_local)),
+        inv_deg_local=torch.from_numpy(1.0 / deg_local),
+        rsqrt_deg_ext=torch.from_numpy(1.0 / np.sqrt(deg_ext)),
+        row_order=torch.from_numpy(row_order),
+        t_row_order=torch.from_numpy(t_row_order),
+        halo_ids=torch.from_numpy(halo_ids),
+        recv_splits=recv_splits,
+        send_idx=send_idx,
+        send_splits=send_splits,
+    )

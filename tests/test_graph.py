@@ -1,0 +1,64 @@
+import numpy as np
+import torch
+
+from roc_amd.graph import (CSRGraph, load_lux, save_lux, synthetic_graph,
+                           synthetic_dataset, build_transpose)
+
+
+def test_synthetic_graph_valid():
+    g = synthetic_graph(100, 1000, seed=3)
+    assert g.num_nodes == 100
+    assert g.rowptr[-1].item() == g.num_edges
+    assert g.colidx.max().item() < 100
+    assert g.colidx.min().item() >= 0
+    # self edges present
+    rp = g.rowptr.numpy()
+    ci = g.colidx.numpy()
+    for v in range(100):
+        assert v in ci[rp[v]:rp[v + 1]]
+    # columns sorted within each row
+    for v in range(100):
+        row = ci[rp[v]:rp[v + 1]]
+        assert np.all(np.diff(row) >= 0)
+
+
+def test_lux_roundtrip(tmp_path):
+    g = synthetic_graph(64, 500, seed=7)
+    p = str(tmp_path / "g.add_self_edge.lux")
+    save_lux(p, g)
+    g2 = load_lux(p)
+    assert g2.num_nodes == g.num_nodes
+    assert g2.num_edges == g.num_edges
+    assert torch.equal(g2.rowptr, g.rowptr)
+    assert torch.equal(g2.colidx, g.colidx)
+
+
+def test_transpose_correct():
+    g = synthetic_graph(50, 400, seed=11)
+    t_rowptr, t_colidx = build_transpose(50, g.rowptr, g.colidx)
+    # dense check: A[v][u] = count of edge u->v in CSR
+    A = np.zeros((50, 50))
+    rp, ci = g.rowptr.numpy(), g.colidx.numpy()
+    for v in range(50):
+        for e in range(rp[v], rp[v + 1]):
+            A[v, ci[e]] += 1
+    At = np.zeros((50, 50))
+    trp, tci = t_rowptr.numpy(), t_colidx.numpy()
+    for u in range(50):
+        for e in range(trp[u], trp[u + 1]):
+            At[u, tci[e]] += 1
+    assert np.array_equal(At, A.T)
+
+
+def test_synthetic_dataset_shapes():
+    g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.1)
+    assert feats.shape[0] == g.num_nodes
+    assert labels.shape[0] == g.num_nodes
+    assert mask.shape[0] == g.num_nodes
+    assert labels.max().item() < c
+
+
+def test_indegree_clamped():
+    g = synthetic_graph(30, 100, seed=5, add_self_edges=False)
+    deg = g.indegree()
+    assert (deg >= 1.0).all()
