@@ -1,0 +1,143 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: per-epoch training time of a 2-layer GCN on a
+Reddit-shaped synthetic graph (BASELINE.json metric), bf16, N GPUs.
+
+Single GPU:   python bench.py --steps 20 --warmup 5
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One JSON line is printed by rank 0. The timed region is K full training
+epochs (zero-grad + forward + loss + backward + grad all-reduce + Adam),
+bracketed by barrier + torch.cuda.synchronize on both sides; the value is
+the MAX per-epoch time over ranks. Synthetic data (no network for real
+Reddit), random-init weights, bf16 compute with fp32 masters.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from roc_amd import (synthetic_dataset, build_shard, build_model,
+                     AdamOptimizer, Trainer)
+from roc_amd.parallel.partition import edge_balanced_bounds
+
+
+def pad_features(feats: torch.Tensor, mult: int = 8) -> torch.Tensor:
+    """Zero-pad the feature dim to a multiple of `mult` (16-B aligned rows
+    for the bf16 kernels; padded weight rows stay exactly zero under
+    Adam+L2 since their grads are identically zero)."""
+    d = feats.shape[1]
+    pad = (-d) % mult
+    if pad == 0:
+        return feats
+    return torch.nn.functional.pad(feats, (0, pad))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--dataset", default="reddit")
+    ap.add_argument("--model", default="gcn")
+    ap.add_argument("--hidden", type=int, default=256)
+    ap.add_argument("--layers", type=int, default=2)
+    ap.add_argument("--scale", type=float, default=1.0,
+                    help="shrink the synthetic graph (debug only)")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--dropout", type=float, default=0.5)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        device = "cuda:0" if torch.cuda.is_available() else "cpu"
+        if device != "cpu":
+            torch.cuda.set_device(0)
+
+    if rank == 0:
+        print(f"[bench] generating synthetic {args.dataset} "
+              f"(scale={args.scale})...", file=sys.stderr, flush=True)
+    t0 = time.perf_counter()
+    g, feats, labels, mask, num_classes = synthetic_dataset(
+        args.dataset, seed=args.seed, scale=args.scale)
+    feats = pad_features(feats)
+    in_dim = feats.shape[1]
+    bounds = edge_balanced_bounds(g.rowptr, world)
+    shard = build_shard(g, rank, world, bounds)
+    if rank == 0:
+        print(f"[bench] graph ready in {time.perf_counter()-t0:.1f}s: "
+              f"{g.num_nodes} nodes, {g.num_edges} edges, "
+              f"halo={shard.n_halo}", file=sys.stderr, flush=True)
+
+    dims = [in_dim] + [args.hidden] * (args.layers - 1) + [num_classes]
+    model = build_model(args.model, dims, dropout=args.dropout, seed=args.seed)
+    opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4,
+                        decay_rate=0.97, decay_steps=100)
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and device != "cpu") \
+        else torch.float32
+    trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
+                      compute_dtype=dtype, grad_scale=1.0, seed=args.seed)
+
+    def barrier():
+        if world > 1:
+            dist.barrier()
+        trainer.sync()
+
+    for _ in range(args.warmup):
+        trainer.train_epoch()
+    barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.train_epoch()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    per_epoch = elapsed / args.steps
+    if rank == 0:
+        result = {
+            "metric": "per-epoch training time (s), 2-layer GCN on Reddit",
+            "value": per_epoch,
+            "unit": "s/epoch",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": per_epoch * 1e3,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": f"{args.model}-{args.layers}layer-" +
+                         "-".join(str(d) for d in dims),
+                "graph": f"{args.dataset}-synthetic-{g.num_nodes}n-{g.num_edges}e",
+                "global_batch": g.num_nodes,
+                "seq_len": None,
+                "parallelism": f"graph-partition x{world} (halo exchange)",
+            },
+        }
+        print(json.dumps(result), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,40 @@
+// Python bindings for the roc_amd CDNA4 kernel library (roc_amd._C).
+#include <torch/extension.h>
+
+void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
+          torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,
+          c10::optional<torch::Tensor> deg_src,
+          c10::optional<torch::Tensor> row_order);
+void rowscale(torch::Tensor out, torch::Tensor x, torch::Tensor scale);
+void relu_fwd(torch::Tensor out, torch::Tensor x);
+void sigmoid_fwd(torch::Tensor out, torch::Tensor x);
+void relu_bwd(torch::Tensor dx, torch::Tensor dy, torch::Tensor y);
+void sigmoid_bwd(torch::Tensor dx, torch::Tensor dy, torch::Tensor y);
+void ewise_add(torch::Tensor out, torch::Tensor a, torch::Tensor b);
+void ewise_mul(torch::Tensor out, torch::Tensor a, torch::Tensor b);
+void dropout_fwd(torch::Tensor out, torch::Tensor x, double p, int64_t seed,
+                 int64_t offset);
+void softmax_ce(torch::Tensor dl, torch::Tensor metrics, torch::Tensor logits,
+                torch::Tensor labels, torch::Tensor mask, double grad_scale);
+void adam_step(torch::Tensor w, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double alpha, double b1, double b2, double eps,
+               double wd);
+void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu);
+void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "roc_amd hand-written CDNA4 (gfx950) kernels";
+  m.def("spmm", &spmm, "CSR SpMM neighbor aggregation (fused deg-norm)");
+  m.def("rowscale", &rowscale);
+  m.def("relu_fwd", &relu_fwd);
+  m.def("sigmoid_fwd", &sigmoid_fwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("sigmoid_bwd", &sigmoid_bwd);
+  m.def("ewise_add", &ewise_add);
+  m.def("ewise_mul", &ewise_mul);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("softmax_ce", &softmax_ce);
+  m.def("adam_step", &adam_step);
+  m.def("gemm_rr", &gemm_rr, "C = A @ Bt^T (bf16 MFMA, optional fused relu)");
+  m.def("gemm_atb", &gemm_atb, "C += A^T @ B (fp32 split-K accumulate)");
+}

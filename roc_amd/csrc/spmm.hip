@@ -1,0 +1,176 @@
+// CSR SpMM neighbor aggregation for CDNA4 — the hot GNN op.
+//
+// out[v, :] = (deg_dst[v] *) sum_{e in row v} (deg_src[u_e] *) x[u_e, :]
+//
+// MI355X-native design (vs reference `scattergather_kernel.cu:20-76`,
+// which staged the full graph tensor host->FB per task and used
+// cub-scan + LDS atomics):
+//  - features stay HBM/L3-resident; no host staging
+//  - a "row team" of TEAM lanes owns one row at a time; each lane
+//    accumulates EPU output elements in registers (no LDS, no atomics)
+//  - 16-B vectorized gathers (uint4 of 8 bf16 / float4)
+//  - optional fused symmetric degree normalization (deg_src/deg_dst are
+//    precomputed rsqrt factors) — replaces two separate norm passes
+//  - rows are visited in degree-descending order (row_order) so hub rows
+//    start first and the skew tail is hidden (Reddit max degree >> mean)
+//  - wide feature dims are column-tiled across blockIdx.y (edges re-read,
+//    colidx is 4 B vs 16 B * TEAM of feature traffic)
+// Backward runs this same kernel on the transpose CSR (exact on
+// asymmetric graphs; the reference assumed symmetry).
+
+#include "common.h"
+
+namespace {
+
+template <typename T, int TEAM>
+__global__ __launch_bounds__(kBlock) void spmm_kernel(
+    T* __restrict__ out, const T* __restrict__ x,
+    const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
+    const float* __restrict__ deg_dst, const float* __restrict__ deg_src,
+    const int* __restrict__ row_order, int num_rows, int64_t D) {
+  constexpr int EPU = EltTraits<T>::kPerVec;
+  const int tpb = kBlock / TEAM;
+  const int team = blockIdx.x * tpb + (int)threadIdx.x / TEAM;
+  const int lane = (int)threadIdx.x % TEAM;
+  const int nteams = gridDim.x * tpb;
+
+  const int64_t col0 = ((int64_t)blockIdx.y * TEAM + lane) * EPU;
+  const bool full = (col0 + EPU) <= D;
+  const int nvalid = full ? EPU : (col0 < D ? (int)(D - col0) : 0);
+
+  for (int ri = team; ri < num_rows; ri += nteams) {
+    const int row = row_order ? row_order[ri] : ri;
+    const int64_t e0 = rowptr[row];
+    const int64_t e1 = rowptr[row + 1];
+    float acc[EPU];
+#pragma unroll
+    for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+
+    int64_t e = e0;
+    // 2-deep unroll: two independent gathers in flight
+    for (; e + 1 < e1; e += 2) {
+      const int u0 = colidx[e];
+      const int u1 = colidx[e + 1];
+      const T* r0 = x + (int64_t)u0 * D + col0;
+      const T* r1 = x + (int64_t)u1 * D + col0;
+      float b0[EPU], b1[EPU];
+      if (full) {
+        if constexpr (EPU == 8) { load_bf16x8(r0, b0); load_bf16x8(r1, b1); }
+        else                    { load_f32x4(r0, b0);  load_f32x4(r1, b1); }
+      } else {
+        for (int j = 0; j < nvalid; ++j) { b0[j] = elt_to_f32(r0[j]); }
+        for (int j = 0; j < nvalid; ++j) { b1[j] = elt_to_f32(r1[j]); }
+        for (int j = nvalid; j < EPU; ++j) { b0[j] = 0.f; b1[j] = 0.f; }
+      }
+      if (deg_src) {
+        const float w0 = deg_src[u0];
+        const float w1 = deg_src[u1];
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) acc[j] += w0 * b0[j] + w1 * b1[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) acc[j] += b0[j] + b1[j];
+      }
+    }
+    if (e < e1) {
+      const int u0 = colidx[e];
+      const T* r0 = x + (int64_t)u0 * D + col0;
+      float b0[EPU];
+      if (full) {
+        if constexpr (EPU == 8) load_bf16x8(r0, b0); else load_f32x4(r0, b0);
+      } else {
+        for (int j = 0; j < nvalid; ++j) b0[j] = elt_to_f32(r0[j]);
+        for (int j = nvalid; j < EPU; ++j) b0[j] = 0.f;
+      }
+      const float w0 = deg_src ? deg_src[u0] : 1.f;
+#pragma unroll
+      for (int j = 0; j < EPU; ++j) acc[j] += w0 * b0[j];
+    }
+
+    if (deg_dst) {
+      const float s = deg_dst[row];
+#pragma unroll
+      for (int j = 0; j < EPU; ++j) acc[j] *= s;
+    }
+    T* o = out + (int64_t)row * D + col0;
+    if (full) {
+      if constexpr (EPU == 8) store_bf16x8(o, acc); else store_f32x4(o, acc);
+    } else {
+      for (int j = 0; j < nvalid; ++j) f32_to_elt(acc[j], o + j);
+    }
+  }
+}
+
+template <typename T>
+void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
+                 const float* deg_dst, const float* deg_src,
+                 const int* row_order, int num_rows, int64_t D,
+                 hipStream_t stream) {
+  constexpr int EPU = EltTraits<T>::kPerVec;
+  const int64_t units = (D + EPU - 1) / EPU;
+  int team = 8;
+  while (team < units && team < 64) team *= 2;
+  const int col_tiles = (int)((units + team - 1) / team);
+  const int tpb = kBlock / team;
+  dim3 grid(roc_grid_1d(num_rows, tpb, 8192), col_tiles);
+  switch (team) {
+    case 8:
+      hipLaunchKernelGGL((spmm_kernel<T, 8>), grid, dim3(kBlock), 0, stream,
+                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
+                         num_rows, D);
+      break;
+    case 16:
+      hipLaunchKernelGGL((spmm_kernel<T, 16>), grid, dim3(kBlock), 0, stream,
+                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
+                         num_rows, D);
+      break;
+    case 32:
+      hipLaunchKernelGGL((spmm_kernel<T, 32>), grid, dim3(kBlock), 0, stream,
+                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
+                         num_rows, D);
+      break;
+    default:
+      hipLaunchKernelGGL((spmm_kernel<T, 64>), grid, dim3(kBlock), 0, stream,
+                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
+                         num_rows, D);
+  }
+}
+
+}  // namespace
+
+void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
+          torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,
+          c10::optional<torch::Tensor> deg_src,
+          c10::optional<torch::Tensor> row_order) {
+  ROC_CHECK_DEV_CONT(out);
+  ROC_CHECK_DEV_CONT(x);
+  ROC_CHECK_DEV_CONT(rowptr);
+  ROC_CHECK_DEV_CONT(colidx);
+  TORCH_CHECK(rowptr.scalar_type() == torch::kInt64, "rowptr must be int64");
+  TORCH_CHECK(colidx.scalar_type() == torch::kInt32, "colidx must be int32");
+  TORCH_CHECK(out.scalar_type() == x.scalar_type(), "dtype mismatch");
+  const int num_rows = (int)out.size(0);
+  const int64_t D = out.size(1);
+  TORCH_CHECK(x.size(1) == D, "feature dim mismatch");
+  TORCH_CHECK(rowptr.size(0) == num_rows + 1, "rowptr size mismatch");
+  const float* dd =
+      deg_dst.has_value() ? deg_dst->data_ptr<float>() : nullptr;
+  const float* ds =
+      deg_src.has_value() ? deg_src->data_ptr<float>() : nullptr;
+  const int* ro =
+      row_order.has_value() ? row_order->data_ptr<int>() : nullptr;
+  auto stream = roc_stream();
+  if (x.scalar_type() == torch::kBFloat16) {
+    launch_spmm<unsigned short>(
+        (unsigned short*)out.data_ptr(), (const unsigned short*)x.data_ptr(),
+        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
+        num_rows, D, stream);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    launch_spmm<float>(out.data_ptr<float>(), x.data_ptr<float>(),
+                       rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd,
+                       ds, ro, num_rows, D, stream);
+  } else {
+    TORCH_CHECK(false, "spmm: unsupported dtype (bf16/f32 only)");
+  }
+  ROC_HIP_CHECK(hipGetLastError());
+}

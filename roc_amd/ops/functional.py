@@ -155,9 +155,10 @@ class _Linear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, act):
         if _hip(x):
-            wc = w.to(x.dtype)
+            # gemm_rr takes B pre-transposed: Bt = w^T [out,in]
+            wt = w.t().contiguous().to(x.dtype)
             y = torch.empty(x.shape[0], w.shape[1], dtype=x.dtype, device=x.device)
-            _C.gemm_rr(y, x, wc, act == "relu")
+            _C.gemm_rr(y, x, wt, act == "relu")
             if act == "sigmoid":
                 _C.sigmoid_fwd(y, y)
         else:
@@ -186,9 +187,10 @@ class _Linear(torch.autograd.Function):
                 dy = dym
             dw = torch.zeros_like(w)  # fp32 [in, out]
             _C.gemm_atb(dw, x, dy)
-            wt = w.t().contiguous().to(dy.dtype)
+            # dx = dy @ w^T: gemm_rr's Bt = (w^T)^T = w as stored [in,out]
+            wc = w.contiguous().to(dy.dtype)
             dx = torch.empty_like(x)
-            _C.gemm_rr(dx, dy, wt, False)
+            _C.gemm_rr(dx, dy, wc, False)
         else:
             if act == "relu":
                 dy = ref.relu_grad(dy, y)

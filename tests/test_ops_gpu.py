@@ -1,0 +1,254 @@
+"""GPU kernel numerics: every CDNA4 kernel vs the plain-PyTorch fp32
+reference of the same op (bf16 paths compare against fp32 references
+computed FROM the bf16-rounded inputs, so only accumulation error
+remains)."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from roc_amd.graph import synthetic_graph, MASK_TRAIN
+from roc_amd.parallel.partition import build_shard
+from roc_amd.ops import functional as F
+from roc_amd.ops import reference as ref
+
+DEV = "cuda:0"
+
+
+def _ext():
+    from roc_amd import _C
+    return _C
+
+
+@pytest.fixture(scope="module")
+def graph_small():
+    g = synthetic_graph(3000, 60000, seed=5)
+    shard = build_shard(g, 0, 1)
+    return g, shard
+
+
+# ---------------------------------------------------------------------------
+# SpMM
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("D", [256, 48, 41, 602])
+def test_spmm_fp32_matches_cpu(graph_small, D):
+    g, shard = graph_small
+    torch.manual_seed(0)
+    x = torch.randn(g.num_nodes, D)
+    want = ref.spmm(x, g.rowptr, g.colidx, g.num_nodes)
+    shd = shard.to(DEV)
+    got = F.scatter_gather(x.to(DEV), shd).cpu()
+    assert torch.allclose(got, want, rtol=1e-4, atol=1e-3), \
+        (got - want).abs().max()
+
+
+@pytest.mark.parametrize("D", [256, 41])
+def test_spmm_bf16(graph_small, D):
+    g, shard = graph_small
+    torch.manual_seed(1)
+    x = torch.randn(g.num_nodes, D).to(torch.bfloat16)
+    want = ref.spmm(x.float(), g.rowptr, g.colidx, g.num_nodes)
+    shd = shard.to(DEV)
+    got = F.scatter_gather(x.to(DEV), shd).cpu().float()
+    # bf16 output rounding of fp32 accumulation
+    tol = want.abs().max().item() * 2 ** -7
+    assert torch.allclose(got, want, atol=tol, rtol=0.02), \
+        (got - want).abs().max()
+
+
+def test_spmm_fused_norm(graph_small):
+    g, shard = graph_small
+    torch.manual_seed(2)
+    x = torch.randn(g.num_nodes, 64)
+    shd = shard.to(DEV)
+    got = F.scatter_gather(x.to(DEV), shd, normalize=True).cpu()
+    want = F.scatter_gather(x, shard, normalize=True)
+    assert torch.allclose(got, want, rtol=1e-3, atol=1e-4), \
+        (got - want).abs().max()
+
+
+def test_spmm_backward_gpu(graph_small):
+    g, shard = graph_small
+    torch.manual_seed(3)
+    shd = shard.to(DEV)
+    x = torch.randn(g.num_nodes, 32, device=DEV, requires_grad=True)
+    out = F.scatter_gather(x, shd)
+    gy = torch.randn_like(out)
+    out.backward(gy)
+    xc = x.detach().cpu().clone().requires_grad_(True)
+    out_c = F.scatter_gather(xc, shard)
+    out_c.backward(gy.cpu())
+    assert torch.allclose(x.grad.cpu(), xc.grad, rtol=1e-4, atol=1e-3)
+
+
+# ---------------------------------------------------------------------------
+# GEMM (MFMA)
+# ---------------------------------------------------------------------------
+
+def test_gemm_rr_identity_asymmetric():
+    # A = I: C must equal B exactly (transpose-detecting per CDNA guide §3)
+    n = 64
+    A = torch.eye(n, dtype=torch.bfloat16, device=DEV)
+    B = torch.arange(n * 48, dtype=torch.float32, device=DEV).reshape(n, 48)
+    B = ((B % 13) - 6).to(torch.bfloat16)  # asymmetric, small ints (exact)
+    C = torch.empty(n, 48, dtype=torch.bfloat16, device=DEV)
+    _ext().gemm_rr(C, A, B.t().contiguous(), False)
+    assert torch.equal(C, B), (C.float() - B.float()).abs().max()
+
+
+@pytest.mark.parametrize("M,K,N", [(300, 608, 256), (257, 256, 41),
+                                   (128, 41, 602), (1000, 256, 128),
+                                   (64, 32, 64)])
+def test_gemm_rr_random(M, K, N):
+    torch.manual_seed(4)
+    A = torch.randn(M, K).to(torch.bfloat16)
+    B = torch.randn(K, N).to(torch.bfloat16)
+    want = A.float() @ B.float()
+    C = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    _ext().gemm_rr(C, A.to(DEV), B.t().contiguous().to(DEV), False)
+    got = C.cpu().float()
+    tol = want.abs().max().item() * 2 ** -7 + 1e-3
+    assert torch.allclose(got, want, atol=tol, rtol=0.05), \
+        (got - want).abs().max()
+
+
+def test_gemm_rr_fused_relu():
+    torch.manual_seed(5)
+    A = torch.randn(200, 64).to(torch.bfloat16)
+    B = torch.randn(64, 96).to(torch.bfloat16)
+    want = torch.relu(A.float() @ B.float())
+    C = torch.empty(200, 96, dtype=torch.bfloat16, device=DEV)
+    _ext().gemm_rr(C, A.to(DEV), B.t().contiguous().to(DEV), True)
+    got = C.cpu().float()
+    tol = want.abs().max().item() * 2 ** -7 + 1e-3
+    assert torch.allclose(got, want, atol=tol, rtol=0.05)
+
+
+@pytest.mark.parametrize("R,Ka,N", [(5000, 608, 256), (3000, 256, 41),
+                                    (1000, 41, 64)])
+def test_gemm_atb(R, Ka, N):
+    torch.manual_seed(6)
+    A = torch.randn(R, Ka).to(torch.bfloat16)
+    B = (torch.randn(R, N) * 0.1).to(torch.bfloat16)
+    want = A.float().t() @ B.float()
+    C = torch.zeros(Ka, N, dtype=torch.float32, device=DEV)
+    _ext().gemm_atb(C, A.to(DEV), B.to(DEV))
+    got = C.cpu()
+    tol = want.abs().max().item() * 2 ** -7 + 1e-2
+    assert torch.allclose(got, want, atol=tol, rtol=0.05), \
+        (got - want).abs().max()
+
+
+def test_linear_autograd_gpu():
+    torch.manual_seed(7)
+    x = torch.randn(500, 64, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    w = torch.nn.Parameter(torch.randn(64, 32, device=DEV))
+    y = F.linear(x, w, activation="relu")
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xc = x.detach().float().cpu().requires_grad_(True)
+    wc = torch.nn.Parameter(w.detach().cpu())
+    yc = F.linear(xc, wc, activation="relu")
+    yc.backward(gy.float().cpu())
+    assert torch.allclose(y.float().cpu(), yc, atol=0.5, rtol=0.05)
+    assert torch.allclose(w.grad.cpu(), wc.grad, atol=2.0, rtol=0.05), \
+        (w.grad.cpu() - wc.grad).abs().max()
+    assert torch.allclose(x.grad.float().cpu(), xc.grad, atol=0.5, rtol=0.05)
+
+
+# ---------------------------------------------------------------------------
+# elementwise / rowscale / dropout / softmax / adam
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_elementwise_gpu(dtype):
+    torch.manual_seed(8)
+    a = torch.randn(999, 7).to(dtype).to(DEV)
+    b = torch.randn(999, 7).to(dtype).to(DEV)
+    out = torch.empty_like(a)
+    _ext().ewise_add(out, a, b)
+    assert torch.allclose(out.float().cpu(), (a + b).float().cpu(), atol=1e-2)
+    _ext().ewise_mul(out, a, b)
+    assert torch.allclose(out.float().cpu(), (a * b).float().cpu(), atol=1e-2)
+    _ext().relu_fwd(out, a)
+    assert torch.equal(out.cpu(), torch.relu(a).cpu())
+    _ext().sigmoid_fwd(out, a)
+    assert torch.allclose(out.float().cpu(), torch.sigmoid(a.float()).cpu(),
+                          atol=1e-2)
+
+
+def test_rowscale_gpu():
+    x = torch.randn(123, 37, device=DEV)
+    s = torch.rand(123, device=DEV) + 0.5
+    out = torch.empty_like(x)
+    _ext().rowscale(out, x, s)
+    assert torch.allclose(out.cpu(), (x * s.unsqueeze(1)).cpu(), atol=1e-6)
+
+
+def test_dropout_gpu():
+    x = torch.ones(100000, device=DEV, dtype=torch.bfloat16).reshape(1000, 100)
+    y = torch.empty_like(x)
+    _ext().dropout_fwd(y, x, 0.5, 1234, 7)
+    kept = (y != 0).float().mean().item()
+    assert 0.48 < kept < 0.52
+    nz = y[y != 0].float()
+    assert torch.allclose(nz, torch.full_like(nz, 2.0))
+    # deterministic: same (seed, offset) -> same mask
+    y2 = torch.empty_like(x)
+    _ext().dropout_fwd(y2, x, 0.5, 1234, 7)
+    assert torch.equal(y, y2)
+    # different offset -> different mask
+    _ext().dropout_fwd(y2, x, 0.5, 1234, 8)
+    assert not torch.equal(y, y2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_softmax_ce_gpu(dtype):
+    torch.manual_seed(9)
+    n, c = 2000, 41
+    logits = torch.randn(n, c).to(dtype)
+    labels = torch.randint(0, c, (n,))
+    mask = torch.randint(1, 4, (n,), dtype=torch.int32)
+    loss, metrics = F.softmax_cross_entropy(
+        logits.to(DEV), labels.to(DEV), mask.to(DEV))
+    dl_want, md_want = ref.softmax_cross_entropy(logits.float(), labels, mask)
+    md = F.decode_metrics(metrics)
+    assert md["train_total"] == md_want["train_total"]
+    assert md["train_acc"] == pytest.approx(
+        md_want["train_correct"] / md_want["train_total"], abs=1e-3)
+    assert md["roc_loss"] == pytest.approx(md_want["roc_loss"], rel=3e-2)
+    assert md["ce_loss"] == pytest.approx(
+        md_want["ce_loss"], rel=3e-2)
+
+
+def test_softmax_ce_grad_gpu():
+    torch.manual_seed(10)
+    n, c = 500, 41
+    logits = torch.randn(n, c, device=DEV, requires_grad=True)
+    labels = torch.randint(0, c, (n,), device=DEV)
+    mask = torch.randint(1, 4, (n,), dtype=torch.int32, device=DEV)
+    loss, _ = F.softmax_cross_entropy(logits, labels, mask)
+    loss.backward()
+    lc = logits.detach().cpu().requires_grad_(True)
+    loss_c, _ = F.softmax_cross_entropy(lc, labels.cpu(), mask.cpu())
+    loss_c.backward()
+    assert torch.allclose(logits.grad.cpu(), lc.grad, atol=1e-4)
+
+
+def test_adam_gpu():
+    torch.manual_seed(11)
+    n = 12345
+    w = torch.randn(n)
+    g = torch.randn(n)
+    m = torch.randn(n).abs()
+    v = torch.randn(n).abs()
+    wd, md_, vd = w.to(DEV), m.to(DEV), v.to(DEV)
+    F.adam_step(wd, g.to(DEV), md_, vd, 0.01, 0.9, 0.999, 1e-8, 1e-4)
+    ref.adam_step(w, g, m, v, 0.01, 0.9, 0.999, 1e-8, 1e-4)
+    assert torch.allclose(wd.cpu(), w, atol=1e-6)
+    assert torch.allclose(md_.cpu(), m, atol=1e-6)
+    assert torch.allclose(vd.cpu(), v, atol=1e-6)
