@@ -41,6 +41,12 @@ class Trainer:
         self.epoch = 0
         F.set_dropout_seed(seed + shard.rank * 7919)
         self._flat_grad = None
+        self.tracer = None  # set via enable_tracing()
+
+    def enable_tracing(self):
+        from .utils.trace import Tracer
+        self.tracer = Tracer(device=str(self.device))
+        return self.tracer
 
     # -- gradient all-reduce (flat bucket; weights are small) ---------------
     def _allreduce_grads(self):
@@ -64,11 +70,25 @@ class Trainer:
 
     def train_epoch(self):
         self.model.train()
-        self.optimizer.zero_grad()
-        loss, metrics = self._forward_loss()
-        loss.backward()
-        self._allreduce_grads()
-        self.optimizer.step()
+        if self.tracer is None:
+            self.optimizer.zero_grad()
+            loss, metrics = self._forward_loss()
+            loss.backward()
+            self._allreduce_grads()
+            self.optimizer.step()
+        else:
+            tr = self.tracer
+            with tr.span("zero_grad"):
+                self.optimizer.zero_grad()
+            with tr.span("forward"):
+                loss, metrics = self._forward_loss()
+            with tr.span("backward"):
+                loss.backward()
+            with tr.span("grad_allreduce"):
+                self._allreduce_grads()
+            with tr.span("adam"):
+                self.optimizer.step()
+            tr.next_epoch()
         self.epoch += 1
         return metrics
 

@@ -45,10 +45,13 @@ class AdamOptimizer:
         lr_t = self.current_lr()
         alpha_t = lr_t * math.sqrt(1.0 - self.beta2 ** self.t) / (
             1.0 - self.beta1 ** self.t)
-        for p, m, v in zip(self.params, self.m, self.v):
+        for i, p in enumerate(self.params):
             if p.grad is None:
                 continue
-            F.adam_step(p.data, p.grad.data, m, v, alpha_t,
+            if self.m[i].device != p.device:  # model moved after ctor
+                self.m[i] = self.m[i].to(p.device)
+                self.v[i] = self.v[i].to(p.device)
+            F.adam_step(p.data, p.grad.data, self.m[i], self.v[i], alpha_t,
                         self.beta1, self.beta2, self.eps, self.weight_decay)
 
     def state_dict(self) -> dict:

@@ -155,7 +155,9 @@ def test_linear_autograd_gpu():
     yc = F.linear(xc, wc, activation="relu")
     yc.backward(gy.float().cpu())
     assert torch.allclose(y.float().cpu(), yc, atol=0.5, rtol=0.05)
-    assert torch.allclose(w.grad.cpu(), wc.grad, atol=2.0, rtol=0.05), \
+    # dW reduces K=500 bf16 products (CPU ref uses fp32 dy): scale-relative tol
+    tol = 0.05 * wc.grad.abs().max().item()
+    assert torch.allclose(w.grad.cpu(), wc.grad, atol=tol, rtol=0.05), \
         (w.grad.cpu() - wc.grad).abs().max()
     assert torch.allclose(x.grad.float().cpu(), xc.grad, atol=0.5, rtol=0.05)
 

@@ -1,0 +1,170 @@
+#!/usr/bin/env python3
+"""Training driver — the framework's CLI (reference `gnn.cc:25-179`).
+
+Single process:
+  python train.py --dataset cora-synthetic --layers 1433-16-7 --epochs 200
+Multi GPU (one process per GPU, RCCL):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+      --master-addr 127.0.0.1 train.py --dataset reddit-synthetic ...
+
+Reference flag parity: --file (.lux dataset prefix), --layers D0-D1-...-C,
+--epochs, --lr, --weight-decay, --decay-rate, --decay-steps, --dropout,
+--seed; metrics printed every 5 epochs (`gnn.cc:107-110`). New:
+--model gcn|sage|gin, --dtype, --checkpoint/--resume, --trace.
+"""
+import argparse
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+from roc_amd import (synthetic_dataset, build_shard, build_model,
+                     AdamOptimizer, Trainer)
+from roc_amd.graph import (load_lux, load_features, load_labels, load_mask,
+                           DATASET_SHAPES)
+from roc_amd.parallel.partition import edge_balanced_bounds
+from roc_amd.utils import save_checkpoint, load_checkpoint
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--file", default=None,
+                    help=".lux dataset prefix (expects <p>.add_self_edge.lux,"
+                         " <p>.feats.csv/.bin, <p>.label, <p>.mask)")
+    ap.add_argument("--dataset", default="cora-synthetic",
+                    help="<name>-synthetic for generated data; see "
+                         "roc_amd.graph.DATASET_SHAPES")
+    ap.add_argument("--scale", type=float, default=1.0)
+    ap.add_argument("--layers", default=None,
+                    help="dash-separated dims incl. input and classes, "
+                         "e.g. 602-256-41")
+    ap.add_argument("--hidden", type=int, default=256)
+    ap.add_argument("--num-layers", type=int, default=2)
+    ap.add_argument("--model", default="gcn", choices=["gcn", "sage", "gin"])
+    ap.add_argument("--epochs", "-e", type=int, default=100)
+    ap.add_argument("--lr", type=float, default=0.01)
+    ap.add_argument("--weight-decay", "--wd", type=float, default=1e-4)
+    ap.add_argument("--decay-rate", type=float, default=0.97)
+    ap.add_argument("--decay-steps", type=int, default=100)
+    ap.add_argument("--dropout", type=float, default=0.5)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--dtype", default="auto", choices=["auto", "bf16", "fp32"])
+    ap.add_argument("--loss", default="mean", choices=["mean", "sum"],
+                    help="sum replicates the reference's unscaled gradient")
+    ap.add_argument("--eval-every", type=int, default=5)
+    ap.add_argument("--residual", action="store_true")
+    ap.add_argument("--checkpoint", default=None)
+    ap.add_argument("--checkpoint-every", type=int, default=0)
+    ap.add_argument("--resume", default=None)
+    ap.add_argument("--trace", default=None,
+                    help="write a chrome trace JSON here at the end")
+    ap.add_argument("-v", "--verbose", action="store_true")
+    return ap.parse_args()
+
+
+def load_dataset(args):
+    if args.file:
+        g = load_lux(args.file + ".add_self_edge.lux")
+        # infer in_dim from the layers flag (reference requires -layers)
+        assert args.layers, "--file requires --layers D0-...-C"
+        dims = [int(d) for d in args.layers.split("-")]
+        feats = load_features(args.file, g.num_nodes, dims[0])
+        labels = load_labels(args.file + ".label", g.num_nodes)
+        mask = load_mask(args.file + ".mask", g.num_nodes)
+        return g, feats, labels, mask, dims[-1]
+    name = args.dataset.replace("-synthetic", "")
+    assert name in DATASET_SHAPES, f"unknown dataset {name}"
+    return synthetic_dataset(name, seed=args.seed, scale=args.scale)
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    on_gpu = torch.cuda.is_available()
+    if world > 1:
+        dist.init_process_group("nccl" if on_gpu else "gloo",
+                                rank=rank, world_size=world)
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        if on_gpu:
+            torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}" if on_gpu else "cpu"
+    else:
+        device = "cuda:0" if on_gpu else "cpu"
+
+    g, feats, labels, mask, num_classes = load_dataset(args)
+    # pad feature dim for 16-B-aligned bf16 rows (zero cols; exact math)
+    pad = (-feats.shape[1]) % 8
+    if pad and on_gpu:
+        feats = torch.nn.functional.pad(feats, (0, pad))
+
+    if args.layers:
+        dims = [int(d) for d in args.layers.split("-")]
+        dims[0] = feats.shape[1]
+    else:
+        dims = [feats.shape[1]] + [args.hidden] * (args.num_layers - 1) \
+            + [num_classes]
+
+    bounds = edge_balanced_bounds(g.rowptr, world)
+    shard = build_shard(g, rank, world, bounds)
+    mkw = {"residual": args.residual} if args.model == "gcn" else {}
+    model = build_model(args.model, dims, dropout=args.dropout,
+                        seed=args.seed, **mkw)
+    opt = AdamOptimizer(model.parameters(), lr=args.lr,
+                        weight_decay=args.weight_decay,
+                        decay_rate=args.decay_rate,
+                        decay_steps=args.decay_steps)
+    dtype = torch.float32
+    if args.dtype == "bf16" or (args.dtype == "auto" and on_gpu):
+        dtype = torch.bfloat16
+    n_train = int((mask == 1).sum())
+    gs = 1.0 if args.loss == "sum" else 1.0 / max(n_train, 1)
+    trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
+                      compute_dtype=dtype, grad_scale=gs, seed=args.seed)
+    if args.trace:
+        trainer.enable_tracing()
+    if args.resume:
+        load_checkpoint(args.resume, trainer)
+        if rank == 0:
+            print(f"resumed from {args.resume} at epoch {trainer.epoch}")
+
+    if rank == 0:
+        print(f"[config] model={args.model} dims={dims} nodes={g.num_nodes} "
+              f"edges={g.num_edges} world={world} device={device} "
+              f"dtype={dtype} lr={args.lr} wd={args.weight_decay} "
+              f"dropout={args.dropout}", flush=True)
+
+    t_start = time.perf_counter()
+    while trainer.epoch < args.epochs:
+        trainer.train_epoch()
+        ep = trainer.epoch
+        if args.eval_every and ep % args.eval_every == 0:
+            md = trainer.evaluate()
+            if rank == 0:
+                dt = time.perf_counter() - t_start
+                print(f"epoch {ep:5d}  loss {md['ce_loss']:.4f}  "
+                      f"train {md['train_acc']:.4f}  val {md['val_acc']:.4f}"
+                      f"  test {md['test_acc']:.4f}  "
+                      f"[{dt / ep * 1e3:.1f} ms/epoch]", flush=True)
+        if (args.checkpoint and args.checkpoint_every
+                and ep % args.checkpoint_every == 0):
+            save_checkpoint(args.checkpoint, trainer)
+
+    if args.checkpoint:
+        save_checkpoint(args.checkpoint, trainer)
+    if args.trace and trainer.tracer is not None:
+        trainer.tracer.dump_chrome(args.trace, rank)
+        if rank == 0:
+            summary = trainer.tracer.summarize()
+            total = sum(summary.values())
+            print("[trace] phase totals (ms): " + ", ".join(
+                f"{k}={v:.1f}" for k, v in sorted(summary.items())) +
+                f"  (sum {total:.1f})")
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
