@@ -48,6 +48,12 @@ class Trainer:
         self.tracer = Tracer(device=str(self.device))
         return self.tracer
 
+    def enable_offload(self, min_bytes: int = 1 << 22):
+        """Host-DRAM activation offload (capacity tier, config #5)."""
+        from .memory import ActivationOffload
+        self.offload = ActivationOffload(min_bytes=min_bytes)
+        return self.offload
+
     # -- gradient all-reduce (flat bucket; weights are small) ---------------
     def _allreduce_grads(self):
         if self.shard.world_size == 1 or not dist.is_initialized():
@@ -63,7 +69,14 @@ class Trainer:
             g.copy_(flat[off:off + n].view_as(g))
             off += n
 
+    offload = None  # set via enable_offload()
+
     def _forward_loss(self):
+        if self.offload is not None:
+            with self.offload:
+                logits = self.model(self.x, self.shard, self.group)
+                return F.softmax_cross_entropy(
+                    logits, self.labels, self.mask, self.grad_scale)
         logits = self.model(self.x, self.shard, self.group)
         return F.softmax_cross_entropy(
             logits, self.labels, self.mask, self.grad_scale)

@@ -1,0 +1,96 @@
+"""Host-DRAM activation offload (capacity tier).
+
+For graphs whose activations exceed the 288 GB HBM budget (papers100M-
+scale, BASELINE.json config #5), saved-for-backward activations are
+copied to pinned host memory on a dedicated HIP stream as soon as the
+producing op finishes, and copied back when autograd needs them. The
+reference kept ALL activations in pinned host (ZC) memory and paid
+H2D+D2H on every op (`types.cu:22-32`); here HBM residency is the
+default and host offload is opt-in spill, double-buffered via
+hipMemcpyAsync on a side stream.
+
+Implementation: torch.autograd.graph.saved_tensors_hooks + a pinned
+buffer pool. `record_stream` keeps the GPU source alive until the D2H
+copy completes.
+"""
+from __future__ import annotations
+
+import torch
+
+
+class ActivationOffload:
+    """Context manager: offload saved activations >= min_bytes to host.
+
+    with ActivationOffload(min_bytes=1 << 22):
+        loss = model(...); loss.backward()
+    """
+
+    def __init__(self, min_bytes: int = 1 << 22, enabled: bool = True):
+        self.min_bytes = min_bytes
+        self.enabled = enabled and torch.cuda.is_available()
+        self.stream = torch.cuda.Stream() if self.enabled else None
+        self._pool = {}   # (shape, dtype) -> list of free pinned tensors
+        self._stats = {"offloaded_bytes": 0, "tensors": 0}
+        self._ctx = None
+
+    # -- pinned buffer pool -------------------------------------------------
+    def _acquire(self, t: torch.Tensor) -> torch.Tensor:
+        key = (tuple(t.shape), t.dtype)
+        free = self._pool.get(key)
+        if free:
+            return free.pop()
+        return torch.empty(t.shape, dtype=t.dtype, device="cpu",
+                           pin_memory=True)
+
+    def _release(self, cpu: torch.Tensor):
+        key = (tuple(cpu.shape), cpu.dtype)
+        self._pool.setdefault(key, []).append(cpu)
+
+    # -- hooks ---------------------------------------------------------------
+    def _pack(self, t: torch.Tensor):
+        if (not self.enabled or not t.is_cuda
+                or t.numel() * t.element_size() < self.min_bytes
+                or (t.is_leaf and t.requires_grad)):  # keep params resident
+            return t
+        cur = torch.cuda.current_stream(t.device)
+        cpu = self._acquire(t)
+        self.stream.wait_stream(cur)          # producer finished
+        with torch.cuda.stream(self.stream):
+            cpu.copy_(t, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record(self.stream)
+        t.record_stream(self.stream)          # allocator: defer reuse
+        self._stats["offloaded_bytes"] += t.numel() * t.element_size()
+        self._stats["tensors"] += 1
+        return ("roc_offloaded", cpu, t.device, ev)
+
+    def _unpack(self, packed):
+        if not isinstance(packed, tuple) or not packed or \
+                packed[0] != "roc_offloaded":
+            return packed
+        _, cpu, device, ev = packed
+        cur = torch.cuda.current_stream(device)
+        with torch.cuda.stream(self.stream):
+            ev.wait(self.stream)              # D2H done before reuse/H2D
+            gpu = cpu.to(device, non_blocking=True)
+            ev2 = torch.cuda.Event()
+            ev2.record(self.stream)
+        cur.wait_event(ev2)
+        gpu.record_stream(cur)
+        self._release(cpu)
+        return gpu
+
+    def __enter__(self):
+        self._ctx = torch.autograd.graph.saved_tensors_hooks(
+            self._pack, self._unpack)
+        self._ctx.__enter__()
+        return self
+
+    def __exit__(self, *a):
+        self._ctx.__exit__(*a)
+        self._ctx = None
+        return False
+
+    @property
+    def stats(self):
+        return dict(self._stats)
