@@ -81,14 +81,18 @@ def main():
               f"{g.num_nodes} nodes, {g.num_edges} edges, "
               f"halo={shard.n_halo}", file=sys.stderr, flush=True)
 
-    dims = [in_dim] + [args.hidden] * (args.layers - 1) + [num_classes]
+    # pad the class dim to a multiple of 8 too (16-B-aligned bf16 rows for
+    # the logits SpMM/GEMM); softmax runs over the true num_classes
+    c_pad = num_classes + ((-num_classes) % 8)
+    dims = [in_dim] + [args.hidden] * (args.layers - 1) + [c_pad]
     model = build_model(args.model, dims, dropout=args.dropout, seed=args.seed)
     opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4,
                         decay_rate=0.97, decay_steps=100)
     dtype = torch.bfloat16 if (args.dtype == "bf16" and device != "cpu") \
         else torch.float32
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
-                      compute_dtype=dtype, grad_scale=1.0, seed=args.seed)
+                      compute_dtype=dtype, grad_scale=1.0, seed=args.seed,
+                      num_classes=num_classes)
 
     def barrier():
         if world > 1:

@@ -15,12 +15,15 @@ void ewise_mul(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void dropout_fwd(torch::Tensor out, torch::Tensor x, double p, int64_t seed,
                  int64_t offset);
 void softmax_ce(torch::Tensor dl, torch::Tensor metrics, torch::Tensor logits,
-                torch::Tensor labels, torch::Tensor mask, double grad_scale);
+                torch::Tensor labels, torch::Tensor mask, double grad_scale,
+                int64_t num_classes);
 void adam_step(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double alpha, double b1, double b2, double eps,
                double wd);
-void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu);
+void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu,
+             c10::optional<torch::Tensor> row_scale);
 void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B);
+void register_graph_cpu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "roc_amd hand-written CDNA4 (gfx950) kernels";
@@ -33,8 +36,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ewise_add", &ewise_add);
   m.def("ewise_mul", &ewise_mul);
   m.def("dropout_fwd", &dropout_fwd);
-  m.def("softmax_ce", &softmax_ce);
+  m.def("softmax_ce", &softmax_ce, pybind11::arg("dl"), pybind11::arg("metrics"),
+        pybind11::arg("logits"), pybind11::arg("labels"), pybind11::arg("mask"),
+        pybind11::arg("grad_scale"), pybind11::arg("num_classes") = -1);
   m.def("adam_step", &adam_step);
-  m.def("gemm_rr", &gemm_rr, "C = A @ Bt^T (bf16 MFMA, optional fused relu)");
+  m.def("gemm_rr", &gemm_rr, "C = A @ Bt^T (bf16 MFMA, fused relu/rowscale)",
+        pybind11::arg("C"), pybind11::arg("A"), pybind11::arg("Bt"),
+        pybind11::arg("relu"), pybind11::arg("row_scale") = pybind11::none());
   m.def("gemm_atb", &gemm_atb, "C += A^T @ B (fp32 split-K accumulate)");
+  register_graph_cpu(m);
 }

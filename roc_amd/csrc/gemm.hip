@@ -62,7 +62,8 @@ __device__ __forceinline__ void stage_tile(
 template <int BN, bool RELU, bool ALIGNED_A, bool ALIGNED_B>
 __global__ __launch_bounds__(kBlock) void gemm_rr_kernel(
     unsigned short* __restrict__ C, const unsigned short* __restrict__ A,
-    const unsigned short* __restrict__ Bt, int M, int N, int K) {
+    const unsigned short* __restrict__ Bt, const float* __restrict__ row_scale,
+    int M, int N, int K) {
   constexpr int NFRAG = BN / 16;
   __shared__ unsigned short a_lds[BM * PADK];
   __shared__ unsigned short b_lds[BN * PADK];
@@ -115,6 +116,7 @@ __global__ __launch_bounds__(kBlock) void gemm_rr_kernel(
         const int col = n_blk + ni * 16 + l15;
         if (row < M && col < N) {
           float v = acc[mi][ni][j];
+          if (row_scale) v *= row_scale[row];  // fused rowwise scale
           if (RELU) v = v > 0.f ? v : 0.f;
           f32_to_elt(v, &C[(int64_t)row * N + col]);
         }
@@ -125,13 +127,13 @@ __global__ __launch_bounds__(kBlock) void gemm_rr_kernel(
 
 template <int BN>
 void launch_rr(unsigned short* C, const unsigned short* A,
-               const unsigned short* Bt, int M, int N, int K, bool relu,
-               hipStream_t s) {
+               const unsigned short* Bt, const float* row_scale, int M, int N,
+               int K, bool relu, hipStream_t s) {
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
   const bool al = (K % 8) == 0;  // both A and Bt have row stride K
 #define ROC_RR_CASE(RELU_, ALA)                                              \
   hipLaunchKernelGGL((gemm_rr_kernel<BN, RELU_, ALA, ALA>), grid,            \
-                     dim3(kBlock), 0, s, C, A, Bt, M, N, K)
+                     dim3(kBlock), 0, s, C, A, Bt, row_scale, M, N, K)
   if (relu) { if (al) ROC_RR_CASE(true, true); else ROC_RR_CASE(true, false); }
   else      { if (al) ROC_RR_CASE(false, true); else ROC_RR_CASE(false, false); }
 #undef ROC_RR_CASE
@@ -224,7 +226,8 @@ __global__ __launch_bounds__(kBlock) void gemm_atb_kernel(
 
 }  // namespace
 
-void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu) {
+void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu,
+             c10::optional<torch::Tensor> row_scale) {
   ROC_CHECK_DEV_CONT(C);
   ROC_CHECK_DEV_CONT(A);
   ROC_CHECK_DEV_CONT(Bt);
@@ -240,10 +243,16 @@ void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu) {
   auto* c = (unsigned short*)C.data_ptr();
   auto* a = (const unsigned short*)A.data_ptr();
   auto* b = (const unsigned short*)Bt.data_ptr();
+  const float* rs = nullptr;
+  if (row_scale.has_value()) {
+    TORCH_CHECK(row_scale->scalar_type() == torch::kFloat32 &&
+                row_scale->numel() == M, "row_scale must be fp32 [M]");
+    rs = row_scale->data_ptr<float>();
+  }
   if (N > 64)
-    launch_rr<128>(c, a, b, M, N, K, relu, s);
+    launch_rr<128>(c, a, b, rs, M, N, K, relu, s);
   else
-    launch_rr<64>(c, a, b, M, N, K, relu, s);
+    launch_rr<64>(c, a, b, rs, M, N, K, relu, s);
   ROC_HIP_CHECK(hipGetLastError());
 }
 

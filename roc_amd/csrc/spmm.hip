@@ -47,32 +47,41 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
     for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
 
     int64_t e = e0;
-    // 2-deep unroll: two independent gathers in flight
-    for (; e + 1 < e1; e += 2) {
-      const int u0 = colidx[e];
-      const int u1 = colidx[e + 1];
-      const T* r0 = x + (int64_t)u0 * D + col0;
-      const T* r1 = x + (int64_t)u1 * D + col0;
-      float b0[EPU], b1[EPU];
+    // 4-deep unroll: four independent gathers in flight per team
+    for (; e + 3 < e1; e += 4) {
+      int u[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) u[q] = colidx[e + q];
+      float b[4][EPU];
       if (full) {
-        if constexpr (EPU == 8) { load_bf16x8(r0, b0); load_bf16x8(r1, b1); }
-        else                    { load_f32x4(r0, b0);  load_f32x4(r1, b1); }
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          const T* r = x + (int64_t)u[q] * D + col0;
+          if constexpr (EPU == 8) load_bf16x8(r, b[q]); else load_f32x4(r, b[q]);
+        }
       } else {
-        for (int j = 0; j < nvalid; ++j) { b0[j] = elt_to_f32(r0[j]); }
-        for (int j = 0; j < nvalid; ++j) { b1[j] = elt_to_f32(r1[j]); }
-        for (int j = nvalid; j < EPU; ++j) { b0[j] = 0.f; b1[j] = 0.f; }
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          const T* r = x + (int64_t)u[q] * D + col0;
+          for (int j = 0; j < nvalid; ++j) b[q][j] = elt_to_f32(r[j]);
+          for (int j = nvalid; j < EPU; ++j) b[q][j] = 0.f;
+        }
       }
       if (deg_src) {
-        const float w0 = deg_src[u0];
-        const float w1 = deg_src[u1];
+        float w[4];
 #pragma unroll
-        for (int j = 0; j < EPU; ++j) acc[j] += w0 * b0[j] + w1 * b1[j];
+        for (int q = 0; q < 4; ++q) w[q] = deg_src[u[q]];
+#pragma unroll
+        for (int j = 0; j < EPU; ++j)
+          acc[j] += w[0] * b[0][j] + w[1] * b[1][j] + w[2] * b[2][j] +
+                    w[3] * b[3][j];
       } else {
 #pragma unroll
-        for (int j = 0; j < EPU; ++j) acc[j] += b0[j] + b1[j];
+        for (int j = 0; j < EPU; ++j)
+          acc[j] += (b[0][j] + b[1][j]) + (b[2][j] + b[3][j]);
       }
     }
-    if (e < e1) {
+    for (; e < e1; ++e) {
       const int u0 = colidx[e];
       const T* r0 = x + (int64_t)u0 * D + col0;
       float b0[EPU];

@@ -26,7 +26,8 @@ class Trainer:
                  labels: torch.Tensor, mask: torch.Tensor,
                  optimizer: AdamOptimizer, device="cpu",
                  compute_dtype: torch.dtype = torch.float32,
-                 grad_scale: float = 1.0, group=None, seed: int = 1):
+                 grad_scale: float = 1.0, group=None, seed: int = 1,
+                 num_classes=None):
         self.model = model.to(device)
         self.shard = shard.to(device)
         self.device = torch.device(device)
@@ -38,6 +39,7 @@ class Trainer:
         self.labels = labels[lo:hi].to(device=device).contiguous()
         self.mask = mask[lo:hi].to(device=device, dtype=torch.int32).contiguous()
         self.optimizer = optimizer
+        self.num_classes = num_classes  # true classes if logits are padded
         self.epoch = 0
         F.set_dropout_seed(seed + shard.rank * 7919)
         self._flat_grad = None
@@ -76,10 +78,12 @@ class Trainer:
             with self.offload:
                 logits = self.model(self.x, self.shard, self.group)
                 return F.softmax_cross_entropy(
-                    logits, self.labels, self.mask, self.grad_scale)
+                    logits, self.labels, self.mask, self.grad_scale,
+                    self.num_classes)
         logits = self.model(self.x, self.shard, self.group)
         return F.softmax_cross_entropy(
-            logits, self.labels, self.mask, self.grad_scale)
+            logits, self.labels, self.mask, self.grad_scale,
+            self.num_classes)
 
     def train_epoch(self):
         self.model.train()
@@ -110,7 +114,8 @@ class Trainer:
         self.model.eval()
         logits = self.model(self.x, self.shard, self.group)
         _, metrics = F.softmax_cross_entropy(
-            logits, self.labels, self.mask, self.grad_scale)
+            logits, self.labels, self.mask, self.grad_scale,
+            self.num_classes)
         if self.shard.world_size > 1 and dist.is_initialized():
             dist.all_reduce(metrics, group=self.group)
         return F.decode_metrics(metrics)

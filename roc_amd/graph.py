@@ -96,7 +96,17 @@ def build_transpose(num_nodes: int, rowptr: torch.Tensor, colidx: torch.Tensor):
 
     Works for rectangular local views too (rowptr rows < colidx id space):
     pass num_nodes = size of the *column* id space.
+
+    Uses the native O(E) counting-sort (roc_amd._C) when built; falls back
+    to numpy argsort (identical, deterministic output).
     """
+    try:
+        from roc_amd import _C
+        t_rowptr, t_colidx = _C.csr_transpose(
+            num_nodes, rowptr.contiguous(), colidx.contiguous())
+        return t_rowptr, t_colidx
+    except ImportError:
+        pass
     nr = rowptr.numel() - 1
     rp = rowptr.numpy()
     ci = colidx.numpy()

@@ -39,11 +39,17 @@ class GCN(torch.nn.Module):
         nlayers = len(self.weights)
         for i, w in enumerate(self.weights):
             h = F.dropout(x, self.p, self.training)
-            h = F.linear(h, w)
-            h = halo_exchange(h, shard, group)
             if self.fused_norm:
-                h = F.scatter_gather(h, shard, normalize=True)
+                # source-side D^-1/2 rides the GEMM epilogue (owner rank
+                # scales its rows BEFORE the halo exchange); dst-side
+                # D^-1/2 rides the SpMM store. No per-edge degree gather.
+                h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
+                h = halo_exchange(h, shard, group)
+                h = F.scatter_gather(h, shard,
+                                     dst_scale=shard.rsqrt_deg_local)
             else:
+                h = F.linear(h, w)
+                h = halo_exchange(h, shard, group)
                 h = F.indegree_norm(
                     F.scatter_gather(F.degree_scale(h, shard.rsqrt_deg_ext), shard),
                     shard)

@@ -83,16 +83,20 @@ class _SpMM(torch.autograd.Function):
         return (dx,) + (None,) * 10
 
 
-def scatter_gather(x, shard, normalize: bool = False):
+def scatter_gather(x, shard, normalize: bool = False, dst_scale=None,
+                   src_scale=None):
     """Neighbor sum-aggregation over the shard's local CSR.
 
     x: [n_ext, D] halo-extended features. Returns [n_local, D].
     normalize=True fuses the symmetric D^-1/2 A D^-1/2 GCN normalization
     (the reference composes indegree_norm -> scatter_gather -> indegree_norm,
     `gnn.cc:82-84`; the fused kernel reads each feature once instead).
+    dst_scale/src_scale: explicit per-row fp32 factors. The fast GCN path
+    pre-scales sources in the linear epilogue and passes only dst_scale
+    here (saves one gather per edge).
     """
-    deg_dst = shard.rsqrt_deg_local if normalize else None
-    deg_src = shard.rsqrt_deg_ext if normalize else None
+    deg_dst = shard.rsqrt_deg_local if normalize else dst_scale
+    deg_src = shard.rsqrt_deg_ext if normalize else src_scale
     return _SpMM.apply(
         x, shard.rowptr, shard.colidx, shard.t_rowptr, shard.t_colidx,
         shard.n_local, shard.n_ext, deg_dst, deg_src,
@@ -153,27 +157,29 @@ class _Linear(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, w, act):
+    def forward(ctx, x, w, act, row_scale):
         if _hip(x):
             # gemm_rr takes B pre-transposed: Bt = w^T [out,in]
             wt = w.t().contiguous().to(x.dtype)
             y = torch.empty(x.shape[0], w.shape[1], dtype=x.dtype, device=x.device)
-            _C.gemm_rr(y, x, wt, act == "relu")
+            _C.gemm_rr(y, x, wt, act == "relu", row_scale)
             if act == "sigmoid":
                 _C.sigmoid_fwd(y, y)
         else:
             y = ref.linear(x, w)
+            if row_scale is not None:
+                y = y * row_scale.unsqueeze(1).to(y.dtype)
             if act == "relu":
                 y = ref.relu(y)
             elif act == "sigmoid":
                 y = ref.sigmoid(y)
-        ctx.save_for_backward(x, w, y if act else None)
+        ctx.save_for_backward(x, w, y if act else None, row_scale)
         ctx.act = act
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w, y = ctx.saved_tensors
+        x, w, y, row_scale = ctx.saved_tensors
         act = ctx.act
         dy = dy.contiguous()
         if _hip(dy):
@@ -184,6 +190,10 @@ class _Linear(torch.autograd.Function):
             elif act == "sigmoid":
                 dym = torch.empty_like(dy)
                 _C.sigmoid_bwd(dym, dy, y)
+                dy = dym
+            if row_scale is not None:
+                dym = torch.empty_like(dy)
+                _C.rowscale(dym, dy, row_scale)
                 dy = dym
             dw = torch.zeros_like(w)  # fp32 [in, out]
             _C.gemm_atb(dw, x, dy)
@@ -196,13 +206,17 @@ class _Linear(torch.autograd.Function):
                 dy = ref.relu_grad(dy, y)
             elif act == "sigmoid":
                 dy = ref.sigmoid_grad(dy, y)
+            if row_scale is not None:
+                dy = dy * row_scale.unsqueeze(1).to(dy.dtype)
             dw = (x.to(torch.float32).t() @ dy.to(torch.float32))
             dx = (dy @ w.t().to(dy.dtype))
-        return dx, dw, None
+        return dx, dw, None, None
 
 
-def linear(x, w, activation: Optional[str] = None):
-    return _Linear.apply(x, w, activation)
+def linear(x, w, activation: Optional[str] = None, row_scale=None):
+    """y = act((x @ w) * row_scale). row_scale rides the GEMM epilogue
+    (used for the GCN source-degree pre-normalization)."""
+    return _Linear.apply(x, w, activation, row_scale)
 
 
 # ---------------------------------------------------------------------------
@@ -352,14 +366,16 @@ class _SoftmaxCE(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, logits, labels, mask, grad_scale):
+    def forward(ctx, logits, labels, mask, grad_scale, num_classes):
         if _hip(logits):
             dl = torch.empty_like(logits)
             metrics = torch.zeros(8, dtype=torch.float32, device=logits.device)
-            _C.softmax_ce(dl, metrics, logits, labels, mask, grad_scale)
+            _C.softmax_ce(dl, metrics, logits, labels, mask, grad_scale,
+                          num_classes or -1)
             loss = metrics[1].clone()
         else:
-            dl, md = ref.softmax_cross_entropy(logits, labels, mask, grad_scale)
+            dl, md = ref.softmax_cross_entropy(logits, labels, mask,
+                                               grad_scale, num_classes)
             metrics = torch.tensor(
                 [md["roc_loss"], md["ce_loss"] * max(md["train_total"], 1),
                  md["train_correct"], md["train_total"],
@@ -374,11 +390,14 @@ class _SoftmaxCE(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss, _dmetrics):
         (dl,) = ctx.saved_tensors
-        return dl * dloss, None, None, None
+        return dl * dloss, None, None, None, None
 
 
-def softmax_cross_entropy(logits, labels, mask, grad_scale: float = 1.0):
-    return _SoftmaxCE.apply(logits, labels, mask, grad_scale)
+def softmax_cross_entropy(logits, labels, mask, grad_scale: float = 1.0,
+                          num_classes: Optional[int] = None):
+    """num_classes < logits width runs softmax over the first num_classes
+    columns only (class-dim padding for 16-B-aligned rows)."""
+    return _SoftmaxCE.apply(logits, labels, mask, grad_scale, num_classes)
 
 
 def decode_metrics(metrics: torch.Tensor) -> dict:

@@ -70,6 +70,7 @@ def softmax_cross_entropy(
     labels: torch.Tensor,
     mask: torch.Tensor,
     grad_scale: float = 1.0,
+    num_classes=None,
 ):
     """Fused softmax -> CE gradient with train-mask zeroing + metrics.
 
@@ -77,8 +78,18 @@ def softmax_cross_entropy(
       dlogit = softmax(logit) - onehot(label), zeroed where mask != Train
       roc_loss = sum over train rows of (1 - p_true)   (the reference's "loss")
     We additionally report true mean cross-entropy over train rows.
+    num_classes < logits width: softmax over the first num_classes cols
+    only (padded class dim); pad columns get zero gradient.
     Returns (dlogits, metrics dict).
     """
+    stride = logits.shape[1]
+    C = num_classes or stride
+    if C < stride:
+        dl_full = torch.zeros_like(logits)
+        dl, md = softmax_cross_entropy(
+            logits[:, :C].contiguous(), labels, mask, grad_scale)
+        dl_full[:, :C] = dl
+        return dl_full, md
     lf = logits.to(torch.float32)
     p = torch.softmax(lf, dim=1)
     n, c = lf.shape

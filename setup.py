@@ -31,13 +31,14 @@ setup(
             name="roc_amd._C",
             sources=sources(),
             extra_compile_args={
-                "cxx": ["-O3", "-std=c++17"],
+                "cxx": ["-O3", "-std=c++17", "-fopenmp"],
                 "nvcc": [
                     "-O3",
                     "-std=c++17",
                     "--offload-arch=gfx950",
                 ],
             },
+            extra_link_args=["-fopenmp"],
         )
     ],
     cmdclass={"build_ext": BuildExtension},
