@@ -8,19 +8,82 @@
 //  - features stay HBM/L3-resident; no host staging
 //  - a "row team" of TEAM lanes owns one row at a time; each lane
 //    accumulates EPU output elements in registers (no LDS, no atomics)
-//  - 16-B vectorized gathers (uint4 of 8 bf16 / float4)
-//  - optional fused symmetric degree normalization (deg_src/deg_dst are
-//    precomputed rsqrt factors) — replaces two separate norm passes
+//  - 16-B vectorized gathers (uint4 of 8 bf16 / float4), 4-deep unrolled
+//  - optional fused degree scaling (deg_* are precomputed factors); the
+//    fast GCN path pre-scales sources in the GEMM epilogue so only the
+//    cheap per-row dst factor remains here
 //  - rows are visited in degree-descending order (row_order) so hub rows
 //    start first and the skew tail is hidden (Reddit max degree >> mean)
-//  - wide feature dims are column-tiled across blockIdx.y (edges re-read,
-//    colidx is 4 B vs 16 B * TEAM of feature traffic)
+//  - wide feature dims are column-tiled across blockIdx.y
+//
+// The vector path and the (rare) partial-unit tail path are fully
+// separated: the hot path's per-slot buffers are only ever statically
+// indexed, so nothing lands in scratch (a shared runtime-indexed array
+// cost 144 B/lane of scratch and 3x wall time — see
+// profiles/r02 notes / CDNA guide §5.4 rule 20).
 // Backward runs this same kernel on the transpose CSR (exact on
 // asymmetric graphs; the reference assumed symmetry).
 
 #include "common.h"
 
 namespace {
+
+template <typename T, int EPU>
+__device__ __forceinline__ void load_unit(const T* p, float* dst) {
+  if constexpr (EPU == 8) load_bf16x8(p, dst); else load_f32x4(p, dst);
+}
+
+// hot path: this lane's 16-B unit is entirely inside D
+template <typename T, int EPU>
+__device__ __forceinline__ void row_accum_vec(
+    float* __restrict__ acc, const T* __restrict__ x, int64_t D, int64_t col0,
+    const int* __restrict__ colidx, const float* __restrict__ deg_src,
+    int64_t e0, int64_t e1) {
+  int64_t e = e0;
+  for (; e + 3 < e1; e += 4) {
+    const int u0 = colidx[e], u1 = colidx[e + 1];
+    const int u2 = colidx[e + 2], u3 = colidx[e + 3];
+    float b0[EPU], b1[EPU], b2[EPU], b3[EPU];
+    load_unit<T, EPU>(x + (int64_t)u0 * D + col0, b0);
+    load_unit<T, EPU>(x + (int64_t)u1 * D + col0, b1);
+    load_unit<T, EPU>(x + (int64_t)u2 * D + col0, b2);
+    load_unit<T, EPU>(x + (int64_t)u3 * D + col0, b3);
+    if (deg_src) {
+      const float w0 = deg_src[u0], w1 = deg_src[u1];
+      const float w2 = deg_src[u2], w3 = deg_src[u3];
+#pragma unroll
+      for (int j = 0; j < EPU; ++j)
+        acc[j] += w0 * b0[j] + w1 * b1[j] + w2 * b2[j] + w3 * b3[j];
+    } else {
+#pragma unroll
+      for (int j = 0; j < EPU; ++j)
+        acc[j] += (b0[j] + b1[j]) + (b2[j] + b3[j]);
+    }
+  }
+  for (; e < e1; ++e) {
+    const int u0 = colidx[e];
+    float b0[EPU];
+    load_unit<T, EPU>(x + (int64_t)u0 * D + col0, b0);
+    const float w0 = deg_src ? deg_src[u0] : 1.f;
+#pragma unroll
+    for (int j = 0; j < EPU; ++j) acc[j] += w0 * b0[j];
+  }
+}
+
+// tail path: unit straddles D (only when D % EPU != 0; engine pads dims
+// to 8 so this is cold). Scalar loads, dynamic-bound loops.
+template <typename T, int EPU>
+__device__ void row_accum_tail(
+    float* __restrict__ acc, const T* __restrict__ x, int64_t D, int64_t col0,
+    int nvalid, const int* __restrict__ colidx,
+    const float* __restrict__ deg_src, int64_t e0, int64_t e1) {
+  for (int64_t e = e0; e < e1; ++e) {
+    const int u0 = colidx[e];
+    const T* r = x + (int64_t)u0 * D + col0;
+    const float w0 = deg_src ? deg_src[u0] : 1.f;
+    for (int j = 0; j < nvalid; ++j) acc[j] += w0 * elt_to_f32(r[j]);
+  }
+}
 
 template <typename T, int TEAM>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
@@ -38,74 +101,34 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
   const bool full = (col0 + EPU) <= D;
   const int nvalid = full ? EPU : (col0 < D ? (int)(D - col0) : 0);
 
-  for (int ri = team; ri < num_rows; ri += nteams) {
-    const int row = row_order ? row_order[ri] : ri;
-    const int64_t e0 = rowptr[row];
-    const int64_t e1 = rowptr[row + 1];
-    float acc[EPU];
+  if (full) {
+    for (int ri = team; ri < num_rows; ri += nteams) {
+      const int row = row_order ? row_order[ri] : ri;
+      const int64_t e0 = rowptr[row];
+      const int64_t e1 = rowptr[row + 1];
+      float acc[EPU];
 #pragma unroll
-    for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
-
-    int64_t e = e0;
-    // 4-deep unroll: four independent gathers in flight per team
-    for (; e + 3 < e1; e += 4) {
-      int u[4];
+      for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+      row_accum_vec<T, EPU>(acc, x, D, col0, colidx, deg_src, e0, e1);
+      if (deg_dst) {
+        const float s = deg_dst[row];
 #pragma unroll
-      for (int q = 0; q < 4; ++q) u[q] = colidx[e + q];
-      float b[4][EPU];
-      if (full) {
-#pragma unroll
-        for (int q = 0; q < 4; ++q) {
-          const T* r = x + (int64_t)u[q] * D + col0;
-          if constexpr (EPU == 8) load_bf16x8(r, b[q]); else load_f32x4(r, b[q]);
-        }
-      } else {
-#pragma unroll
-        for (int q = 0; q < 4; ++q) {
-          const T* r = x + (int64_t)u[q] * D + col0;
-          for (int j = 0; j < nvalid; ++j) b[q][j] = elt_to_f32(r[j]);
-          for (int j = nvalid; j < EPU; ++j) b[q][j] = 0.f;
-        }
+        for (int j = 0; j < EPU; ++j) acc[j] *= s;
       }
-      if (deg_src) {
-        float w[4];
-#pragma unroll
-        for (int q = 0; q < 4; ++q) w[q] = deg_src[u[q]];
-#pragma unroll
-        for (int j = 0; j < EPU; ++j)
-          acc[j] += w[0] * b[0][j] + w[1] * b[1][j] + w[2] * b[2][j] +
-                    w[3] * b[3][j];
-      } else {
-#pragma unroll
-        for (int j = 0; j < EPU; ++j)
-          acc[j] += (b[0][j] + b[1][j]) + (b[2][j] + b[3][j]);
-      }
-    }
-    for (; e < e1; ++e) {
-      const int u0 = colidx[e];
-      const T* r0 = x + (int64_t)u0 * D + col0;
-      float b0[EPU];
-      if (full) {
-        if constexpr (EPU == 8) load_bf16x8(r0, b0); else load_f32x4(r0, b0);
-      } else {
-        for (int j = 0; j < nvalid; ++j) b0[j] = elt_to_f32(r0[j]);
-        for (int j = nvalid; j < EPU; ++j) b0[j] = 0.f;
-      }
-      const float w0 = deg_src ? deg_src[u0] : 1.f;
-#pragma unroll
-      for (int j = 0; j < EPU; ++j) acc[j] += w0 * b0[j];
-    }
-
-    if (deg_dst) {
-      const float s = deg_dst[row];
-#pragma unroll
-      for (int j = 0; j < EPU; ++j) acc[j] *= s;
-    }
-    T* o = out + (int64_t)row * D + col0;
-    if (full) {
+      T* o = out + (int64_t)row * D + col0;
       if constexpr (EPU == 8) store_bf16x8(o, acc); else store_f32x4(o, acc);
-    } else {
-      for (int j = 0; j < nvalid; ++j) f32_to_elt(acc[j], o + j);
+    }
+  } else if (nvalid > 0) {
+    for (int ri = team; ri < num_rows; ri += nteams) {
+      const int row = row_order ? row_order[ri] : ri;
+      float acc[EPU];
+#pragma unroll
+      for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+      row_accum_tail<T, EPU>(acc, x, D, col0, nvalid, colidx, deg_src,
+                             rowptr[row], rowptr[row + 1]);
+      const float s = deg_dst ? deg_dst[row] : 1.f;
+      T* o = out + (int64_t)row * D + col0;
+      for (int j = 0; j < nvalid; ++j) f32_to_elt(acc[j] * s, o + j);
     }
   }
 }

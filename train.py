@@ -30,6 +30,8 @@ from roc_amd.utils import save_checkpoint, load_checkpoint
 
 def parse_args():
     ap = argparse.ArgumentParser()
+    ap.add_argument("--config", default=None,
+                    help="YAML config (CLI flags override; see configs/)")
     ap.add_argument("--file", default=None,
                     help=".lux dataset prefix (expects <p>.add_self_edge.lux,"
                          " <p>.feats.csv/.bin, <p>.label, <p>.mask)")
@@ -60,8 +62,27 @@ def parse_args():
     ap.add_argument("--resume", default=None)
     ap.add_argument("--trace", default=None,
                     help="write a chrome trace JSON here at the end")
+    ap.add_argument("--offload", action="store_true",
+                    help="host-DRAM activation offload (capacity tier)")
     ap.add_argument("-v", "--verbose", action="store_true")
-    return ap.parse_args()
+    args = ap.parse_args()
+    if args.config:
+        import yaml
+        with open(args.config) as f:
+            cfg = yaml.safe_load(f) or {}
+        defaults = {}
+        for k, v in cfg.items():
+            key = k.replace("-", "_")
+            assert hasattr(args, key), f"unknown config key {k}"
+            defaults[key] = v
+        # CLI flags that were explicitly set keep priority
+        import sys as _sys
+        given = {a.split("=")[0].lstrip("-").replace("-", "_")
+                 for a in _sys.argv[1:] if a.startswith("--")}
+        for k, v in defaults.items():
+            if k not in given:
+                setattr(args, k, v)
+    return args
 
 
 def load_dataset(args):
@@ -104,8 +125,11 @@ def main():
         dims = [int(d) for d in args.layers.split("-")]
         dims[0] = feats.shape[1]
     else:
+        c_out = num_classes
+        if on_gpu:  # pad class dim for 16-B-aligned logits rows
+            c_out = num_classes + ((-num_classes) % 8)
         dims = [feats.shape[1]] + [args.hidden] * (args.num_layers - 1) \
-            + [num_classes]
+            + [c_out]
 
     bounds = edge_balanced_bounds(g.rowptr, world)
     shard = build_shard(g, rank, world, bounds)
@@ -122,9 +146,13 @@ def main():
     n_train = int((mask == 1).sum())
     gs = 1.0 if args.loss == "sum" else 1.0 / max(n_train, 1)
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
-                      compute_dtype=dtype, grad_scale=gs, seed=args.seed)
+                      compute_dtype=dtype, grad_scale=gs, seed=args.seed,
+                      num_classes=num_classes if dims[-1] != num_classes
+                      else None)
     if args.trace:
         trainer.enable_tracing()
+    if args.offload:
+        trainer.enable_offload()
     if args.resume:
         load_checkpoint(args.resume, trainer)
         if rank == 0:
