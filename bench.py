@@ -56,13 +56,19 @@ def main():
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    on_gpu = torch.cuda.is_available()
     if world > 1:
-        dist.init_process_group("nccl", rank=rank, world_size=world)
+        # nccl IS RCCL on ROCm; gloo keeps the path testable on CPU boxes
+        dist.init_process_group("nccl" if on_gpu else "gloo",
+                                rank=rank, world_size=world)
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
-        torch.cuda.set_device(local_rank)
-        device = f"cuda:{local_rank}"
+        if on_gpu:
+            torch.cuda.set_device(local_rank)
+            device = f"cuda:{local_rank}"
+        else:
+            device = "cpu"
     else:
-        device = "cuda:0" if torch.cuda.is_available() else "cpu"
+        device = "cuda:0" if on_gpu else "cpu"
         if device != "cpu":
             torch.cuda.set_device(0)
 

@@ -1,0 +1,61 @@
+"""Native C++ graph machinery (roc_amd._C CPU functions) vs numpy."""
+import numpy as np
+import pytest
+import torch
+
+from roc_amd.graph import synthetic_graph
+
+_C = pytest.importorskip("roc_amd._C")
+
+
+def numpy_transpose(num_cols, rowptr, colidx):
+    rp = rowptr.numpy()
+    ci = colidx.numpy()
+    counts = np.bincount(ci, minlength=num_cols).astype(np.int64)
+    t_rowptr = np.zeros(num_cols + 1, dtype=np.int64)
+    np.cumsum(counts, out=t_rowptr[1:])
+    dst = np.repeat(np.arange(rp.shape[0] - 1, dtype=np.int32), np.diff(rp))
+    order = np.argsort(ci, kind="stable")
+    return t_rowptr, dst[order]
+
+
+@pytest.mark.parametrize("n,e", [(100, 900), (500, 8000)])
+def test_csr_transpose_matches_numpy(n, e):
+    g = synthetic_graph(n, e, seed=17)
+    trp, tci = _C.csr_transpose(n, g.rowptr, g.colidx)
+    nrp, nci = numpy_transpose(n, g.rowptr, g.colidx)
+    assert np.array_equal(trp.numpy(), nrp)
+    assert np.array_equal(tci.numpy(), nci)
+
+
+def test_csr_transpose_rectangular():
+    # local view: 10 rows, columns in a 30-id ext space
+    rowptr = torch.tensor([0, 2, 4, 6, 8, 10, 12, 14, 16, 18, 20],
+                          dtype=torch.int64)
+    colidx = torch.arange(20, dtype=torch.int32) % 30
+    trp, tci = _C.csr_transpose(30, rowptr, colidx)
+    assert trp.numel() == 31
+    assert trp[-1].item() == 20
+    nrp, nci = numpy_transpose(30, rowptr, colidx)
+    assert np.array_equal(trp.numpy(), nrp)
+    assert np.array_equal(tci.numpy(), nci)
+
+
+def test_csr_sort_rows():
+    g = synthetic_graph(50, 600, seed=3)
+    ci = g.colidx.clone()
+    perm = torch.randperm(ci.numel())
+    # shuffle within the whole array then re-sort rows
+    shuffled = ci.clone()
+    _C.csr_sort_rows(g.rowptr, shuffled)
+    rp = g.rowptr.numpy()
+    s = shuffled.numpy()
+    for v in range(50):
+        assert np.all(np.diff(s[rp[v]:rp[v + 1]]) >= 0)
+
+
+def test_divergence_guard():
+    from roc_amd.debug import check_metrics, TrainingDiverged
+    check_metrics(torch.ones(8))
+    with pytest.raises(TrainingDiverged):
+        check_metrics(torch.tensor([1.0, float("nan"), 0, 0, 0, 0, 0, 0]))
