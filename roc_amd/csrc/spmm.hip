@@ -34,12 +34,46 @@ __device__ __forceinline__ void load_unit(const T* p, float* dst) {
 }
 
 // hot path: this lane's 16-B unit is entirely inside D
-template <typename T, int EPU>
+template <typename T, int EPU, bool UN8>
 __device__ __forceinline__ void row_accum_vec(
     float* __restrict__ acc, const T* __restrict__ x, int64_t D, int64_t col0,
     const int* __restrict__ colidx, const float* __restrict__ deg_src,
     int64_t e0, int64_t e1) {
   int64_t e = e0;
+  if (UN8) {
+    // 8 independent gathers in flight (two statically-named 4-groups)
+    for (; e + 7 < e1; e += 8) {
+      const int u0 = colidx[e], u1 = colidx[e + 1];
+      const int u2 = colidx[e + 2], u3 = colidx[e + 3];
+      const int u4 = colidx[e + 4], u5 = colidx[e + 5];
+      const int u6 = colidx[e + 6], u7 = colidx[e + 7];
+      float b0[EPU], b1[EPU], b2[EPU], b3[EPU];
+      float b4[EPU], b5[EPU], b6[EPU], b7[EPU];
+      load_unit<T, EPU>(x + (int64_t)u0 * D + col0, b0);
+      load_unit<T, EPU>(x + (int64_t)u1 * D + col0, b1);
+      load_unit<T, EPU>(x + (int64_t)u2 * D + col0, b2);
+      load_unit<T, EPU>(x + (int64_t)u3 * D + col0, b3);
+      load_unit<T, EPU>(x + (int64_t)u4 * D + col0, b4);
+      load_unit<T, EPU>(x + (int64_t)u5 * D + col0, b5);
+      load_unit<T, EPU>(x + (int64_t)u6 * D + col0, b6);
+      load_unit<T, EPU>(x + (int64_t)u7 * D + col0, b7);
+      if (deg_src) {
+        const float w0 = deg_src[u0], w1 = deg_src[u1];
+        const float w2 = deg_src[u2], w3 = deg_src[u3];
+        const float w4 = deg_src[u4], w5 = deg_src[u5];
+        const float w6 = deg_src[u6], w7 = deg_src[u7];
+#pragma unroll
+        for (int j = 0; j < EPU; ++j)
+          acc[j] += (w0 * b0[j] + w1 * b1[j] + w2 * b2[j] + w3 * b3[j]) +
+                    (w4 * b4[j] + w5 * b5[j] + w6 * b6[j] + w7 * b7[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < EPU; ++j)
+          acc[j] += ((b0[j] + b1[j]) + (b2[j] + b3[j])) +
+                    ((b4[j] + b5[j]) + (b6[j] + b7[j]));
+      }
+    }
+  }
   for (; e + 3 < e1; e += 4) {
     const int u0 = colidx[e], u1 = colidx[e + 1];
     const int u2 = colidx[e + 2], u3 = colidx[e + 3];
@@ -85,7 +119,7 @@ __device__ void row_accum_tail(
   }
 }
 
-template <typename T, int TEAM>
+template <typename T, int TEAM, bool UN8>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
     T* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
@@ -109,7 +143,7 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
       float acc[EPU];
 #pragma unroll
       for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
-      row_accum_vec<T, EPU>(acc, x, D, col0, colidx, deg_src, e0, e1);
+      row_accum_vec<T, EPU, UN8>(acc, x, D, col0, colidx, deg_src, e0, e1);
       if (deg_dst) {
         const float s = deg_dst[row];
 #pragma unroll
@@ -145,27 +179,27 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   const int col_tiles = (int)((units + team - 1) / team);
   const int tpb = kBlock / team;
   dim3 grid(roc_grid_1d(num_rows, tpb, 8192), col_tiles);
+  // read per call (cheap vs a ms-scale launch) so A/B harnesses can toggle
+  const char* un8_env = getenv("ROC_SPMM_UNROLL8");
+  const bool un8 = !(un8_env && un8_env[0] == '0');  // default ON
+#define ROC_SPMM_LAUNCH(TEAM_, UN8_)                                        \
+  hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_>), grid, dim3(kBlock), 0,  \
+                     stream, out, x, rowptr, colidx, deg_dst, deg_src,      \
+                     row_order, num_rows, D)
   switch (team) {
     case 8:
-      hipLaunchKernelGGL((spmm_kernel<T, 8>), grid, dim3(kBlock), 0, stream,
-                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
-                         num_rows, D);
+      if (un8) { ROC_SPMM_LAUNCH(8, true); } else { ROC_SPMM_LAUNCH(8, false); }
       break;
     case 16:
-      hipLaunchKernelGGL((spmm_kernel<T, 16>), grid, dim3(kBlock), 0, stream,
-                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
-                         num_rows, D);
+      if (un8) { ROC_SPMM_LAUNCH(16, true); } else { ROC_SPMM_LAUNCH(16, false); }
       break;
     case 32:
-      hipLaunchKernelGGL((spmm_kernel<T, 32>), grid, dim3(kBlock), 0, stream,
-                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
-                         num_rows, D);
+      if (un8) { ROC_SPMM_LAUNCH(32, true); } else { ROC_SPMM_LAUNCH(32, false); }
       break;
     default:
-      hipLaunchKernelGGL((spmm_kernel<T, 64>), grid, dim3(kBlock), 0, stream,
-                         out, x, rowptr, colidx, deg_dst, deg_src, row_order,
-                         num_rows, D);
+      if (un8) { ROC_SPMM_LAUNCH(64, true); } else { ROC_SPMM_LAUNCH(64, false); }
   }
+#undef ROC_SPMM_LAUNCH
 }
 
 }  // namespace

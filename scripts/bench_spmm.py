@@ -1,0 +1,76 @@
+#!/usr/bin/env python3
+"""Standalone SpMM kernel A/B microbench (within-process, interleaved
+rounds per CDNA guide §5.4 rule 24). Run on a GPU box:
+
+  python scripts/bench_spmm.py [--nodes N --edges E --rounds R]
+"""
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def time_variant(fn, rounds, stream=None):
+    s = torch.cuda.Event(enable_timing=True)
+    e = torch.cuda.Event(enable_timing=True)
+    fn()  # warm
+    torch.cuda.synchronize()
+    times = []
+    for _ in range(rounds):
+        s.record()
+        fn()
+        e.record()
+        torch.cuda.synchronize()
+        times.append(s.elapsed_time(e))
+    times.sort()
+    return times[len(times) // 2], times[0]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--nodes", type=int, default=232965)
+    ap.add_argument("--edges", type=int, default=114848857)
+    ap.add_argument("--rounds", type=int, default=5)
+    ap.add_argument("--dims", type=int, nargs="+", default=[48, 256])
+    args = ap.parse_args()
+
+    from roc_amd.graph import synthetic_graph
+    from roc_amd import _C
+
+    g = synthetic_graph(args.nodes, args.edges, seed=1)
+    dev = "cuda:0"
+    rowptr = g.rowptr.to(dev)
+    colidx = g.colidx.to(dev)
+    deg = (g.rowptr[1:] - g.rowptr[:-1]).float().clamp(min=1)
+    rsq = deg.rsqrt().to(dev)
+    row_order = torch.argsort(-deg).int().to(dev)
+
+    for D in args.dims:
+        x = torch.randn(args.nodes, D).to(torch.bfloat16).to(dev)
+        out = torch.empty_like(x)
+        gb = args.edges * D * 2 / 1e9
+        print(f"== D={D}  gather volume {gb:.1f} GB ==")
+        variants = {}
+        for un8 in (0, 1):
+            for use_order in (0, 1):
+                for use_src in (0, 1):
+                    os.environ["ROC_SPMM_UNROLL8"] = str(un8)
+                    ro = row_order if use_order else None
+                    ds = rsq if use_src else None
+
+                    def fn(ro=ro, ds=ds):
+                        _C.spmm(out, x, rowptr, colidx, rsq, ds, ro)
+
+                    med, best = time_variant(fn, args.rounds)
+                    key = f"un8={un8} order={use_order} srcdeg={use_src}"
+                    variants[key] = (med, best)
+                    print(f"  {key}: median {med:8.2f} ms  best {best:8.2f}"
+                          f"  ({gb/med:.2f} TB/s eff)", flush=True)
+    print("done")
+
+
+if __name__ == "__main__":
+    main()
