@@ -13,7 +13,18 @@ namespace {
 __global__ __launch_bounds__(kBlock) void adam_kernel(
     float* __restrict__ w, const float* __restrict__ g,
     float* __restrict__ m, float* __restrict__ v, float alpha, float b1,
-    float b2, float eps, float wd, int64_t n) {
+    float b2, float eps, float wd,
+    const long long* __restrict__ step, float decay_rate, int decay_steps,
+    int64_t n) {
+  // hipGraph-replay support: with `step` given, `alpha` is the BASE lr and
+  // the bias-corrected, decayed step size is derived on device from the
+  // step counter (semantics of optimizer.cc:79-85 + gnn.cc:100-101).
+  if (step) {
+    const float t = (float)(*step);
+    const float lr_t =
+        alpha * __powf(decay_rate, floorf(t / (float)decay_steps));
+    alpha = lr_t * sqrtf(1.f - __powf(b2, t)) / (1.f - __powf(b1, t));
+  }
   const int64_t units = (n + 3) / 4;
   for (int64_t u = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; u < units;
        u += (int64_t)gridDim.x * blockDim.x) {
@@ -52,7 +63,8 @@ __global__ __launch_bounds__(kBlock) void adam_kernel(
 
 void adam_step(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double alpha, double b1, double b2, double eps,
-               double wd) {
+               double wd, c10::optional<torch::Tensor> step,
+               double decay_rate, int64_t decay_steps) {
   ROC_CHECK_DEV_CONT(w);
   ROC_CHECK_DEV_CONT(g);
   ROC_CHECK_DEV_CONT(m);
@@ -62,9 +74,15 @@ void adam_step(torch::Tensor w, torch::Tensor g, torch::Tensor m,
   const int64_t n = w.numel();
   TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n);
   const int grid = roc_grid_1d((n + 3) / 4, kBlock, 1024);
+  const long long* st = nullptr;
+  if (step.has_value()) {
+    TORCH_CHECK(step->scalar_type() == torch::kInt64 && step->is_cuda());
+    st = (const long long*)step->data_ptr<int64_t>();
+  }
   hipLaunchKernelGGL(adam_kernel, dim3(grid), dim3(kBlock), 0, roc_stream(),
                      w.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(), (float)alpha,
-                     (float)b1, (float)b2, (float)eps, (float)wd, n);
+                     (float)b1, (float)b2, (float)eps, (float)wd, st,
+                     (float)decay_rate, (int)decay_steps, n);
   ROC_HIP_CHECK(hipGetLastError());
 }

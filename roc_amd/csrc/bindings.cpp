@@ -13,13 +13,14 @@ void sigmoid_bwd(torch::Tensor dx, torch::Tensor dy, torch::Tensor y);
 void ewise_add(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void ewise_mul(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void dropout_fwd(torch::Tensor out, torch::Tensor x, double p, int64_t seed,
-                 int64_t offset);
+                 int64_t offset, c10::optional<torch::Tensor> counter);
 void softmax_ce(torch::Tensor dl, torch::Tensor metrics, torch::Tensor logits,
                 torch::Tensor labels, torch::Tensor mask, double grad_scale,
                 int64_t num_classes);
 void adam_step(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double alpha, double b1, double b2, double eps,
-               double wd);
+               double wd, c10::optional<torch::Tensor> step,
+               double decay_rate, int64_t decay_steps);
 void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu,
              c10::optional<torch::Tensor> row_scale);
 void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B);
@@ -35,11 +36,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sigmoid_bwd", &sigmoid_bwd);
   m.def("ewise_add", &ewise_add);
   m.def("ewise_mul", &ewise_mul);
-  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_fwd", &dropout_fwd, pybind11::arg("out"), pybind11::arg("x"),
+        pybind11::arg("p"), pybind11::arg("seed"), pybind11::arg("offset"),
+        pybind11::arg("counter") = pybind11::none());
   m.def("softmax_ce", &softmax_ce, pybind11::arg("dl"), pybind11::arg("metrics"),
         pybind11::arg("logits"), pybind11::arg("labels"), pybind11::arg("mask"),
         pybind11::arg("grad_scale"), pybind11::arg("num_classes") = -1);
-  m.def("adam_step", &adam_step);
+  m.def("adam_step", &adam_step, pybind11::arg("w"), pybind11::arg("g"),
+        pybind11::arg("m"), pybind11::arg("v"), pybind11::arg("alpha"),
+        pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
+        pybind11::arg("wd"), pybind11::arg("step") = pybind11::none(),
+        pybind11::arg("decay_rate") = 1.0,
+        pybind11::arg("decay_steps") = 100);
   m.def("gemm_rr", &gemm_rr, "C = A @ Bt^T (bf16 MFMA, fused relu/rowscale)",
         pybind11::arg("C"), pybind11::arg("A"), pybind11::arg("Bt"),
         pybind11::arg("relu"), pybind11::arg("row_scale") = pybind11::none());

@@ -45,11 +45,16 @@ constexpr float kU32ToUnit = 2.3283064365386963e-10f;  // 2^-32
 template <typename T>
 __global__ __launch_bounds__(kBlock) void dropout_kernel(
     T* __restrict__ out, const T* __restrict__ x, float p, float inv_keep,
-    unsigned long long seed, unsigned offset, int64_t n) {
+    unsigned long long seed, unsigned offset,
+    const long long* __restrict__ counter, int64_t n) {
+  // hipGraph-replay support: the epoch index lives on the DEVICE so a
+  // captured graph draws fresh Philox streams on every replay.
+  const unsigned off_eff =
+      counter ? (unsigned)(*counter) * 65536u + offset : offset;
   const int64_t units = (n + 3) / 4;
   for (int64_t u = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; u < units;
        u += (int64_t)gridDim.x * blockDim.x) {
-    const uint4 r = philox4x32_10((unsigned long long)u, offset, seed);
+    const uint4 r = philox4x32_10((unsigned long long)u, off_eff, seed);
     const unsigned rnd[4] = {r.x, r.y, r.z, r.w};
     const int64_t i0 = u * 4;
     const int cnt = (int)min((int64_t)4, n - i0);
@@ -66,7 +71,7 @@ __global__ __launch_bounds__(kBlock) void dropout_kernel(
 }  // namespace
 
 void dropout_fwd(torch::Tensor out, torch::Tensor x, double p, int64_t seed,
-                 int64_t offset) {
+                 int64_t offset, c10::optional<torch::Tensor> counter) {
   ROC_CHECK_DEV_CONT(out);
   ROC_CHECK_DEV_CONT(x);
   TORCH_CHECK(out.scalar_type() == x.scalar_type());
@@ -74,16 +79,22 @@ void dropout_fwd(torch::Tensor out, torch::Tensor x, double p, int64_t seed,
   const int64_t n = x.numel();
   const float inv_keep = (float)(1.0 / (1.0 - p));
   const int grid = roc_grid_1d((n + 3) / 4, kBlock, 2048);
+  const long long* cnt = nullptr;
+  if (counter.has_value()) {
+    TORCH_CHECK(counter->scalar_type() == torch::kInt64 && counter->is_cuda());
+    cnt = (const long long*)counter->data_ptr<int64_t>();
+  }
   auto s = roc_stream();
   if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL((dropout_kernel<unsigned short>), dim3(grid),
                        dim3(kBlock), 0, s, (unsigned short*)out.data_ptr(),
                        (const unsigned short*)x.data_ptr(), (float)p, inv_keep,
-                       (unsigned long long)seed, (unsigned)offset, n);
+                       (unsigned long long)seed, (unsigned)offset, cnt, n);
   } else if (x.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((dropout_kernel<float>), dim3(grid), dim3(kBlock), 0, s,
                        out.data_ptr<float>(), x.data_ptr<float>(), (float)p,
-                       inv_keep, (unsigned long long)seed, (unsigned)offset, n);
+                       inv_keep, (unsigned long long)seed, (unsigned)offset,
+                       cnt, n);
   } else {
     TORCH_CHECK(false, "dropout: unsupported dtype");
   }

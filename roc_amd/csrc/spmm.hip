@@ -182,6 +182,10 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   // read per call (cheap vs a ms-scale launch) so A/B harnesses can toggle
   const char* un8_env = getenv("ROC_SPMM_UNROLL8");
   const bool un8 = !(un8_env && un8_env[0] == '0');  // default ON
+  // measured (scripts/bench_spmm.py, Reddit shape): degree-descending
+  // scheduling wins for wide rows (D=256: -13%) but loses for narrow
+  // ones (D=48: +10% — the indirection costs more than the skew tail)
+  if (team < 16) row_order = nullptr;
 #define ROC_SPMM_LAUNCH(TEAM_, UN8_)                                        \
   hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_>), grid, dim3(kBlock), 0,  \
                      stream, out, x, rowptr, colidx, deg_dst, deg_src,      \

@@ -56,6 +56,65 @@ class Trainer:
         self.offload = ActivationOffload(min_bytes=min_bytes)
         return self.offload
 
+    # -- hipGraph capture of the whole training epoch -----------------------
+    # Launch-bound epochs (small shards / many GPUs) replay one hipGraph
+    # instead of ~40 Python-driven launches. Epoch-varying state lives on
+    # DEVICE: an int64 step counter bumped inside the graph drives the
+    # Philox dropout offsets and the Adam bias-corrected/decayed step size.
+    def enable_graph_capture(self, warmup_epochs: int = 2):
+        assert self.device.type == "cuda", "graph capture needs a GPU"
+        self._step_dev = torch.zeros(1, dtype=torch.int64, device=self.device)
+        F.set_dropout_counter(self._step_dev)
+        self.optimizer.set_device_step(self._step_dev)
+        self._graph = None
+        self._graph_warmup = warmup_epochs
+        self._graph_metrics = None
+        self.use_graph = True
+
+    use_graph = False
+
+    def _epoch_body(self):
+        self._step_dev += 1
+        F.reset_dropout_offset()
+        self.optimizer.zero_grad()
+        loss, metrics = self._forward_loss()
+        loss.backward()
+        self._allreduce_grads()
+        self.optimizer.step()
+        return metrics
+
+    def _graph_epoch(self):
+        # (optimizer.step() inside _epoch_body increments host t itself;
+        #  a replay executes no Python, so replays bump it explicitly)
+        if self._graph is not None:
+            self._graph.replay()
+            self.optimizer.t += 1
+            return self._graph_metrics
+        if self._graph_warmup > 0:
+            # warmup eagerly on a side stream (torch capture protocol)
+            self._graph_warmup -= 1
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                metrics = self._epoch_body()
+            torch.cuda.current_stream().wait_stream(s)
+            return metrics
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._graph_metrics = self._epoch_body()
+            self.optimizer.t -= 1  # capture recorded, did not execute
+            self._graph = g
+        except Exception as e:  # pragma: no cover - capture unsupported
+            torch.cuda.synchronize()
+            print(f"[roc_amd] hipGraph capture failed ({e!r}); "
+                  "falling back to eager", flush=True)
+            self.use_graph = False
+            return self._epoch_body()
+        self._graph.replay()
+        self.optimizer.t += 1
+        return self._graph_metrics
+
     # -- gradient all-reduce (flat bucket; weights are small) ---------------
     def _allreduce_grads(self):
         if self.shard.world_size == 1 or not dist.is_initialized():
@@ -87,6 +146,11 @@ class Trainer:
 
     def train_epoch(self):
         self.model.train()
+        if (self.use_graph and self.device.type == "cuda"
+                and self.tracer is None and self.offload is None):
+            metrics = self._graph_epoch()
+            self.epoch += 1
+            return metrics
         if self.tracer is None:
             self.optimizer.zero_grad()
             loss, metrics = self._forward_loss()
@@ -123,6 +187,46 @@ class Trainer:
     def sync(self):
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
+
+    # -- cost-model repartitioning (the MLSys'20 Roc idea; the reference
+    #    code only has the static edge-balanced split) --------------------
+    def measure_and_rebalance(self, feats, labels, mask, probe_epochs=3):
+        """Measure this rank's epoch time, gather all ranks' times, re-split
+        the vertex range so predicted load matches measured throughput, and
+        rebuild the shard in place. Needs the FULL feats/labels/mask (every
+        rank loads/generates the full dataset at startup).
+        Returns the new bounds."""
+        from .parallel.partition import rebalance_bounds, build_shard
+        t = self.timed_epochs(probe_epochs) / probe_epochs
+        if self.shard.world_size == 1 or not dist.is_initialized():
+            return self.shard.bounds
+        times = torch.zeros(self.shard.world_size)
+        times[self.shard.rank] = t
+        dist.all_reduce(times, group=self.group)
+        # rowptr of the full graph is not kept; derive per-rank edge counts
+        # from the current shard (identical on all ranks via bounds)
+        new_bounds = rebalance_bounds(
+            self._full_rowptr, self.shard.bounds, times.tolist())
+        if new_bounds == self.shard.bounds:
+            return new_bounds
+        new_shard = build_shard(self._full_graph, self.shard.rank,
+                                self.shard.world_size, new_bounds)
+        self.load_shard(new_shard, feats, labels, mask)
+        return new_bounds
+
+    def attach_full_graph(self, g):
+        """Keep a handle to the full CPU graph for repartitioning."""
+        self._full_graph = g
+        self._full_rowptr = g.rowptr
+
+    def load_shard(self, shard, feats, labels, mask):
+        device, dt = self.device, self.dtype
+        self.shard = shard.to(device)
+        lo, hi = shard.lo, shard.hi
+        self.x = feats[lo:hi].to(device=device, dtype=dt).contiguous()
+        self.labels = labels[lo:hi].to(device=device).contiguous()
+        self.mask = mask[lo:hi].to(device=device,
+                                   dtype=torch.int32).contiguous()
 
     def timed_epochs(self, n: int) -> float:
         """Run n training epochs, return wall seconds (caller barriers)."""

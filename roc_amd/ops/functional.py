@@ -309,11 +309,22 @@ def mul(a, b):
 # tensor is ever materialized on GPU)
 # ---------------------------------------------------------------------------
 
-_DROPOUT_STATE = {"seed": 1, "offset": 0}
+_DROPOUT_STATE = {"seed": 1, "offset": 0, "counter": None}
 
 
 def set_dropout_seed(seed: int) -> None:
     _DROPOUT_STATE["seed"] = int(seed)
+    _DROPOUT_STATE["offset"] = 0
+
+
+def set_dropout_counter(counter) -> None:
+    """Device int64[1] epoch counter: makes dropout hipGraph-replayable
+    (offset = counter * 65536 + per-epoch call index)."""
+    _DROPOUT_STATE["counter"] = counter
+    _DROPOUT_STATE["offset"] = 0
+
+
+def reset_dropout_offset() -> None:
     _DROPOUT_STATE["offset"] = 0
 
 
@@ -322,12 +333,12 @@ class _Dropout(torch.autograd.Function):
     def forward(ctx, x, p):
         seed = _DROPOUT_STATE["seed"]
         offset = _DROPOUT_STATE["offset"]
+        counter = _DROPOUT_STATE["counter"]
         _DROPOUT_STATE["offset"] += 1
-        ctx.params = (p, seed, offset)
+        ctx.params = (p, seed, offset, counter)
         if _hip(x):
             y = torch.empty_like(x)
-            _C.dropout_fwd(y, x, p, seed, offset)
-            ctx.cpu_mask = None
+            _C.dropout_fwd(y, x, p, seed, offset, counter)
             return y
         mask = ref.dropout_mask(x.shape, p, seed, offset, x.device)
         ctx.save_for_backward(mask)
@@ -335,11 +346,11 @@ class _Dropout(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        p, seed, offset = ctx.params
+        p, seed, offset, counter = ctx.params
         dy = dy.contiguous()
         if _hip(dy):
             dx = torch.empty_like(dy)
-            _C.dropout_fwd(dx, dy, p, seed, offset)  # same Philox stream
+            _C.dropout_fwd(dx, dy, p, seed, offset, counter)  # same stream
             return dx, None
         (mask,) = ctx.saved_tensors
         return dy * mask.to(dy.dtype) / (1.0 - p), None
@@ -419,8 +430,13 @@ def decode_metrics(metrics: torch.Tensor) -> dict:
 # Adam (not autograd — called by the optimizer)
 # ---------------------------------------------------------------------------
 
-def adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay):
+def adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay,
+              step=None, decay_rate=1.0, decay_steps=100):
+    """step (device int64[1]) switches to the hipGraph-replayable mode:
+    alpha_t is then the BASE lr and the bias-corrected decayed step size
+    is derived on device from the counter."""
     if _hip(w):
-        _C.adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay)
+        _C.adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay,
+                     step, decay_rate, decay_steps)
     else:
         ref.adam_step(w, g, m, v, alpha_t, beta1, beta2, eps, weight_decay)

@@ -30,6 +30,12 @@ class AdamOptimizer:
         self.t = 0
         self.m = [torch.zeros_like(p.data) for p in self.params]
         self.v = [torch.zeros_like(p.data) for p in self.params]
+        self._step_dev = None  # device schedule (hipGraph capture mode)
+
+    def set_device_step(self, step_tensor) -> None:
+        """int64[1] device counter (bumped by the engine inside the
+        captured epoch); the Adam kernel derives alpha_t from it."""
+        self._step_dev = step_tensor
 
     def zero_grad(self) -> None:
         for p in self.params:
@@ -42,9 +48,16 @@ class AdamOptimizer:
 
     def step(self) -> None:
         self.t += 1
-        lr_t = self.current_lr()
-        alpha_t = lr_t * math.sqrt(1.0 - self.beta2 ** self.t) / (
-            1.0 - self.beta1 ** self.t)
+        if self._step_dev is None:
+            lr_t = self.current_lr()
+            alpha_t = lr_t * math.sqrt(1.0 - self.beta2 ** self.t) / (
+                1.0 - self.beta1 ** self.t)
+            step_kw = {}
+        else:
+            alpha_t = self.lr  # device derives decayed, bias-corrected rate
+            step_kw = {"step": self._step_dev,
+                       "decay_rate": self.decay_rate,
+                       "decay_steps": self.decay_steps}
         for i, p in enumerate(self.params):
             if p.grad is None:
                 continue
@@ -52,7 +65,8 @@ class AdamOptimizer:
                 self.m[i] = self.m[i].to(p.device)
                 self.v[i] = self.v[i].to(p.device)
             F.adam_step(p.data, p.grad.data, self.m[i], self.v[i], alpha_t,
-                        self.beta1, self.beta2, self.eps, self.weight_decay)
+                        self.beta1, self.beta2, self.eps, self.weight_decay,
+                        **step_kw)
 
     def state_dict(self) -> dict:
         return {"t": self.t, "m": self.m, "v": self.v,
