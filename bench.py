@@ -52,6 +52,8 @@ def main():
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--dropout", type=float, default=0.5)
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph epoch capture")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -87,9 +89,11 @@ def main():
               f"{g.num_nodes} nodes, {g.num_edges} edges, "
               f"halo={shard.n_halo}", file=sys.stderr, flush=True)
 
-    # pad the class dim to a multiple of 8 too (16-B-aligned bf16 rows for
-    # the logits SpMM/GEMM); softmax runs over the true num_classes
-    c_pad = num_classes + ((-num_classes) % 8)
+    # pad the class dim so logits rows are whole 128-B cachelines (64 bf16):
+    # the logits aggregation is gather-request-bound and 96-B rows straddle
+    # lines (~1.75 requests/edge -> 1; PMC evidence in profiles/).
+    # Softmax runs over the true num_classes; pad cols carry zero grads.
+    c_pad = num_classes + ((-num_classes) % 64)
     dims = [in_dim] + [args.hidden] * (args.layers - 1) + [c_pad]
     model = build_model(args.model, dims, dropout=args.dropout, seed=args.seed)
     opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4,
@@ -99,6 +103,8 @@ def main():
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
                       compute_dtype=dtype, grad_scale=1.0, seed=args.seed,
                       num_classes=num_classes)
+    if device != "cpu" and not args.no_graph:
+        trainer.enable_graph_capture()
 
     def barrier():
         if world > 1:

@@ -27,7 +27,8 @@ def save_checkpoint(path: str, trainer, extra: Optional[dict] = None) -> None:
             "v": [t.cpu() for t in trainer.optimizer.v],
         },
         "epoch": trainer.epoch,
-        "dropout_state": dict(F._DROPOUT_STATE),
+        "dropout_state": {k: v for k, v in F._DROPOUT_STATE.items()
+                          if k != "counter"},
         "bounds": trainer.shard.bounds,
         "world_size": trainer.shard.world_size,
         "torch_rng": torch.get_rng_state(),
@@ -50,6 +51,10 @@ def load_checkpoint(path: str, trainer) -> dict:
     for dst, src in zip(opt.v, state["optim"]["v"]):
         dst.copy_(src.to(dst.device))
     trainer.epoch = state["epoch"]
+    if getattr(trainer, "_step_dev", None) is not None:
+        trainer._step_dev.fill_(opt.t)  # resync device schedule counter
+    counter = F._DROPOUT_STATE.get("counter")
     F._DROPOUT_STATE.update(state["dropout_state"])
+    F._DROPOUT_STATE["counter"] = counter  # device tensor is not persisted
     torch.set_rng_state(state["torch_rng"])
     return state.get("extra", {})

@@ -1,0 +1,73 @@
+"""hipGraph whole-epoch capture: replayed epochs must train like eager
+ones (advancing dropout streams + Adam schedule on device)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from roc_amd import (synthetic_dataset, build_shard, build_model,
+                     AdamOptimizer, Trainer)
+from roc_amd.ops import functional as F
+
+
+def _make(graph_capture, dropout=0.3, seed=11):
+    g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.2, seed=7)
+    pad = (-feats.shape[1]) % 8
+    if pad:
+        feats = torch.nn.functional.pad(feats, (0, pad))
+    shard = build_shard(g, 0, 1)
+    model = build_model("gcn", [feats.shape[1], 64, c], dropout=dropout,
+                        seed=1)
+    opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4,
+                        decay_rate=0.97, decay_steps=100)
+    tr = Trainer(model, shard, feats, labels, mask, opt, device="cuda:0",
+                 compute_dtype=torch.bfloat16, seed=seed)
+    if graph_capture:
+        tr.enable_graph_capture(warmup_epochs=2)
+    return tr
+
+
+def test_capture_matches_eager_no_dropout():
+    F.set_dropout_counter(None)
+    tr_e = _make(False, dropout=0.0)
+    for _ in range(8):
+        tr_e.train_epoch()
+    me = tr_e.evaluate()
+    we = tr_e.model.weights[0].detach().cpu()
+
+    tr_g = _make(True, dropout=0.0)
+    for _ in range(8):
+        tr_g.train_epoch()
+    assert tr_g._graph is not None, "graph was never captured"
+    mg = tr_g.evaluate()
+    wg = tr_g.model.weights[0].detach().cpu()
+    F.set_dropout_counter(None)
+    # identical math modulo split-K atomic ordering noise
+    assert torch.allclose(we, wg, atol=5e-3), (we - wg).abs().max()
+    assert mg["ce_loss"] == pytest.approx(me["ce_loss"], rel=0.02)
+    assert tr_g.optimizer.t == tr_e.optimizer.t == 8
+
+
+def test_capture_dropout_advances():
+    tr = _make(True, dropout=0.5)
+    losses = []
+    for _ in range(10):
+        m = tr.train_epoch()
+    # two consecutive replays must not produce identical metrics
+    torch.cuda.synchronize()
+    m1 = tr.train_epoch().cpu().clone()
+    m2 = tr.train_epoch().cpu().clone()
+    torch.cuda.synchronize()
+    F.set_dropout_counter(None)
+    assert not torch.equal(m1, m2), "dropout mask frozen across replays"
+    assert torch.isfinite(m1).all() and torch.isfinite(m2).all()
+
+
+def test_capture_trains():
+    tr = _make(True, dropout=0.2)
+    m0 = tr.evaluate()
+    for _ in range(40):
+        tr.train_epoch()
+    m1 = tr.evaluate()
+    F.set_dropout_counter(None)
+    assert m1["ce_loss"] < m0["ce_loss"], (m0, m1)

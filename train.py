@@ -126,8 +126,8 @@ def main():
         dims[0] = feats.shape[1]
     else:
         c_out = num_classes
-        if on_gpu:  # pad class dim for 16-B-aligned logits rows
-            c_out = num_classes + ((-num_classes) % 8)
+        if on_gpu:  # pad class dim to whole 128-B logits rows (see bench.py)
+            c_out = num_classes + ((-num_classes) % 64)
         dims = [feats.shape[1]] + [args.hidden] * (args.num_layers - 1) \
             + [c_out]
 
