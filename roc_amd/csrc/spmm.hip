@@ -28,9 +28,27 @@
 
 namespace {
 
-template <typename T, int EPU>
-__device__ __forceinline__ void load_unit(const T* p, float* dst) {
-  if constexpr (EPU == 8) load_bf16x8(p, dst); else load_f32x4(p, dst);
+// raw 16-B gather payload: unpacking is deferred to accumulate time so
+// an in-flight slot costs 4 VGPRs, not 8 (occupancy = latency hiding)
+template <typename T>
+struct RawVec { using type = float4; };
+template <>
+struct RawVec<unsigned short> { using type = uint4; };
+
+__device__ __forceinline__ uint4 load_raw(const unsigned short* p) {
+  return *reinterpret_cast<const uint4*>(p);
+}
+__device__ __forceinline__ float4 load_raw(const float* p) {
+  return *reinterpret_cast<const float4*>(p);
+}
+__device__ __forceinline__ void acc_add(float* acc, const uint4& r, float w) {
+  acc[0] += w * bf16_lo(r.x); acc[1] += w * bf16_hi(r.x);
+  acc[2] += w * bf16_lo(r.y); acc[3] += w * bf16_hi(r.y);
+  acc[4] += w * bf16_lo(r.z); acc[5] += w * bf16_hi(r.z);
+  acc[6] += w * bf16_lo(r.w); acc[7] += w * bf16_hi(r.w);
+}
+__device__ __forceinline__ void acc_add(float* acc, const float4& r, float w) {
+  acc[0] += w * r.x; acc[1] += w * r.y; acc[2] += w * r.z; acc[3] += w * r.w;
 }
 
 // hot path: this lane's 16-B unit is entirely inside D
@@ -39,68 +57,55 @@ __device__ __forceinline__ void row_accum_vec(
     float* __restrict__ acc, const T* __restrict__ x, int64_t D, int64_t col0,
     const int* __restrict__ colidx, const float* __restrict__ deg_src,
     int64_t e0, int64_t e1) {
+  using Raw = typename RawVec<T>::type;
   int64_t e = e0;
   if (UN8) {
-    // 8 independent gathers in flight (two statically-named 4-groups)
+    // 8 independent gathers in flight (raw payloads, unpack at use)
     for (; e + 7 < e1; e += 8) {
       const int u0 = colidx[e], u1 = colidx[e + 1];
       const int u2 = colidx[e + 2], u3 = colidx[e + 3];
       const int u4 = colidx[e + 4], u5 = colidx[e + 5];
       const int u6 = colidx[e + 6], u7 = colidx[e + 7];
-      float b0[EPU], b1[EPU], b2[EPU], b3[EPU];
-      float b4[EPU], b5[EPU], b6[EPU], b7[EPU];
-      load_unit<T, EPU>(x + (int64_t)u0 * D + col0, b0);
-      load_unit<T, EPU>(x + (int64_t)u1 * D + col0, b1);
-      load_unit<T, EPU>(x + (int64_t)u2 * D + col0, b2);
-      load_unit<T, EPU>(x + (int64_t)u3 * D + col0, b3);
-      load_unit<T, EPU>(x + (int64_t)u4 * D + col0, b4);
-      load_unit<T, EPU>(x + (int64_t)u5 * D + col0, b5);
-      load_unit<T, EPU>(x + (int64_t)u6 * D + col0, b6);
-      load_unit<T, EPU>(x + (int64_t)u7 * D + col0, b7);
+      const Raw r0 = load_raw(x + (int64_t)u0 * D + col0);
+      const Raw r1 = load_raw(x + (int64_t)u1 * D + col0);
+      const Raw r2 = load_raw(x + (int64_t)u2 * D + col0);
+      const Raw r3 = load_raw(x + (int64_t)u3 * D + col0);
+      const Raw r4 = load_raw(x + (int64_t)u4 * D + col0);
+      const Raw r5 = load_raw(x + (int64_t)u5 * D + col0);
+      const Raw r6 = load_raw(x + (int64_t)u6 * D + col0);
+      const Raw r7 = load_raw(x + (int64_t)u7 * D + col0);
       if (deg_src) {
-        const float w0 = deg_src[u0], w1 = deg_src[u1];
-        const float w2 = deg_src[u2], w3 = deg_src[u3];
-        const float w4 = deg_src[u4], w5 = deg_src[u5];
-        const float w6 = deg_src[u6], w7 = deg_src[u7];
-#pragma unroll
-        for (int j = 0; j < EPU; ++j)
-          acc[j] += (w0 * b0[j] + w1 * b1[j] + w2 * b2[j] + w3 * b3[j]) +
-                    (w4 * b4[j] + w5 * b5[j] + w6 * b6[j] + w7 * b7[j]);
+        acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
+        acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
+        acc_add(acc, r4, deg_src[u4]); acc_add(acc, r5, deg_src[u5]);
+        acc_add(acc, r6, deg_src[u6]); acc_add(acc, r7, deg_src[u7]);
       } else {
-#pragma unroll
-        for (int j = 0; j < EPU; ++j)
-          acc[j] += ((b0[j] + b1[j]) + (b2[j] + b3[j])) +
-                    ((b4[j] + b5[j]) + (b6[j] + b7[j]));
+        acc_add(acc, r0, 1.f); acc_add(acc, r1, 1.f);
+        acc_add(acc, r2, 1.f); acc_add(acc, r3, 1.f);
+        acc_add(acc, r4, 1.f); acc_add(acc, r5, 1.f);
+        acc_add(acc, r6, 1.f); acc_add(acc, r7, 1.f);
       }
     }
   }
   for (; e + 3 < e1; e += 4) {
     const int u0 = colidx[e], u1 = colidx[e + 1];
     const int u2 = colidx[e + 2], u3 = colidx[e + 3];
-    float b0[EPU], b1[EPU], b2[EPU], b3[EPU];
-    load_unit<T, EPU>(x + (int64_t)u0 * D + col0, b0);
-    load_unit<T, EPU>(x + (int64_t)u1 * D + col0, b1);
-    load_unit<T, EPU>(x + (int64_t)u2 * D + col0, b2);
-    load_unit<T, EPU>(x + (int64_t)u3 * D + col0, b3);
+    const Raw r0 = load_raw(x + (int64_t)u0 * D + col0);
+    const Raw r1 = load_raw(x + (int64_t)u1 * D + col0);
+    const Raw r2 = load_raw(x + (int64_t)u2 * D + col0);
+    const Raw r3 = load_raw(x + (int64_t)u3 * D + col0);
     if (deg_src) {
-      const float w0 = deg_src[u0], w1 = deg_src[u1];
-      const float w2 = deg_src[u2], w3 = deg_src[u3];
-#pragma unroll
-      for (int j = 0; j < EPU; ++j)
-        acc[j] += w0 * b0[j] + w1 * b1[j] + w2 * b2[j] + w3 * b3[j];
+      acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
+      acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
     } else {
-#pragma unroll
-      for (int j = 0; j < EPU; ++j)
-        acc[j] += (b0[j] + b1[j]) + (b2[j] + b3[j]);
+      acc_add(acc, r0, 1.f); acc_add(acc, r1, 1.f);
+      acc_add(acc, r2, 1.f); acc_add(acc, r3, 1.f);
     }
   }
   for (; e < e1; ++e) {
     const int u0 = colidx[e];
-    float b0[EPU];
-    load_unit<T, EPU>(x + (int64_t)u0 * D + col0, b0);
-    const float w0 = deg_src ? deg_src[u0] : 1.f;
-#pragma unroll
-    for (int j = 0; j < EPU; ++j) acc[j] += w0 * b0[j];
+    const Raw r0 = load_raw(x + (int64_t)u0 * D + col0);
+    acc_add(acc, r0, deg_src ? deg_src[u0] : 1.f);
   }
 }
 
