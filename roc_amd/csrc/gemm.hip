@@ -156,9 +156,7 @@ __device__ __forceinline__ void stage_tile_T(
   const int cseg = threadIdx.x % 8;            // 8 elems each
   const int r = threadIdx.x / 8;               // 0..31
   const int gr = r0 + r;
-  float tmp[8];
   short8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-  (void)tmp;
   if (gr < nrows) {
     const unsigned short* p = g + (int64_t)gr * ld + c0 + cseg * 8;
     const int nv = min(8, ncols - (c0 + cseg * 8));
@@ -271,6 +269,8 @@ void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B) {
   const int tiles = ((Ka + BKA - 1) / BKA) * ((N + BNW - 1) / BNW);
   int splitk = 2048 / max(tiles, 1);
   splitk = max(1, min(splitk, (R + RB - 1) / RB));
+  const char* det = getenv("ROC_DETERMINISTIC");
+  if (det && det[0] == '1') splitk = 1;  // bit-reproducible dW (slower)
   int rows_per_split = ((R + splitk - 1) / splitk + RB - 1) / RB * RB;
   splitk = (R + rows_per_split - 1) / rows_per_split;
   dim3 grid((Ka + BKA - 1) / BKA, (N + BNW - 1) / BNW, splitk);
