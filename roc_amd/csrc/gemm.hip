@@ -24,22 +24,22 @@ constexpr int BM = 128;   // M rows per block
 constexpr int BK = 32;    // K per step (= one MFMA K)
 constexpr int PADK = 40;  // LDS row stride in elements (80 B, 16B-aligned)
 
-// stage a [rows x BK] bf16 tile from row-major global (ld = row stride in
-// elements) into LDS[rows][PADK], zero-filling out-of-range. 256 threads.
-// aligned==true: 16-B vector loads (requires ld % 8 == 0).
+// Tile staging is split into a LOAD half (global -> regs, issued before
+// the MFMAs of the previous tile so HBM latency hides under compute) and
+// a WRITE half (regs -> LDS after the barrier). CDNA guide T14.
+// 256 threads; thread -> (row, 16-B segment); zero-fill out-of-range.
 template <int ROWS, bool ALIGNED>
-__device__ __forceinline__ void stage_tile(
-    unsigned short* __restrict__ lds, const unsigned short* __restrict__ g,
-    int row0, int nrows, int64_t ld, int k0, int K) {
-  constexpr int SEGS = BK / 8;          // 16-B segments per row
-  constexpr int THREADS_PER_ROW = SEGS; // 4
+__device__ __forceinline__ void stage_load(
+    short8 (&regs)[ROWS * (BK / 8) / kBlock],
+    const unsigned short* __restrict__ g, int row0, int nrows, int64_t ld,
+    int k0, int K) {
+  constexpr int THREADS_PER_ROW = BK / 8;                  // 4
   constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;  // 64
   const int seg = threadIdx.x % THREADS_PER_ROW;
   const int r_in = threadIdx.x / THREADS_PER_ROW;
 #pragma unroll
   for (int pass = 0; pass < ROWS / ROWS_PER_PASS; ++pass) {
-    const int r = pass * ROWS_PER_PASS + r_in;
-    const int gr = row0 + r;
+    const int gr = row0 + pass * ROWS_PER_PASS + r_in;
     const int gk = k0 + seg * 8;
     short8 v = {0, 0, 0, 0, 0, 0, 0, 0};
     if (gr < nrows) {
@@ -51,7 +51,22 @@ __device__ __forceinline__ void stage_tile(
         for (int j = 0; j < nv; ++j) v[j] = (short)p[j];
       }
     }
-    *reinterpret_cast<short8*>(&lds[r * PADK + seg * 8]) = v;
+    regs[pass] = v;
+  }
+}
+
+template <int ROWS>
+__device__ __forceinline__ void stage_write(
+    unsigned short* __restrict__ lds,
+    const short8 (&regs)[ROWS * (BK / 8) / kBlock]) {
+  constexpr int THREADS_PER_ROW = BK / 8;
+  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  const int seg = threadIdx.x % THREADS_PER_ROW;
+  const int r_in = threadIdx.x / THREADS_PER_ROW;
+#pragma unroll
+  for (int pass = 0; pass < ROWS / ROWS_PER_PASS; ++pass) {
+    const int r = pass * ROWS_PER_PASS + r_in;
+    *reinterpret_cast<short8*>(&lds[r * PADK + seg * 8]) = regs[pass];
   }
 }
 
@@ -65,8 +80,8 @@ __global__ __launch_bounds__(kBlock) void gemm_rr_kernel(
     const unsigned short* __restrict__ Bt, const float* __restrict__ row_scale,
     int M, int N, int K) {
   constexpr int NFRAG = BN / 16;
-  __shared__ unsigned short a_lds[BM * PADK];
-  __shared__ unsigned short b_lds[BN * PADK];
+  __shared__ unsigned short a_lds[2][BM * PADK];
+  __shared__ unsigned short b_lds[2][BN * PADK];
 
   const int m_blk = blockIdx.x * BM;
   const int n_blk = blockIdx.y * BN;
@@ -82,27 +97,42 @@ __global__ __launch_bounds__(kBlock) void gemm_rr_kernel(
 #pragma unroll
     for (int j = 0; j < NFRAG; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  short8 ra[BM * (BK / 8) / kBlock];
+  short8 rb[BN * (BK / 8) / kBlock];
   const int nk = (K + BK - 1) / BK;
+  // prologue: tile 0 into LDS buffer 0
+  stage_load<BM, ALIGNED_A>(ra, A, m_blk, M, K, 0, K);
+  stage_load<BN, ALIGNED_B>(rb, Bt, n_blk, N, K, 0, K);
+  stage_write<BM>(a_lds[0], ra);
+  stage_write<BN>(b_lds[0], rb);
+  __syncthreads();
+  int cur = 0;
   for (int kt = 0; kt < nk; ++kt) {
-    const int k0 = kt * BK;
-    stage_tile<BM, ALIGNED_A>(a_lds, A, m_blk, M, K, k0, K);
-    stage_tile<BN, ALIGNED_B>(b_lds, Bt, n_blk, N, K, k0, K);
-    __syncthreads();
+    // issue next tile's global loads before this tile's MFMAs (T14)
+    if (kt + 1 < nk) {
+      stage_load<BM, ALIGNED_A>(ra, A, m_blk, M, K, (kt + 1) * BK, K);
+      stage_load<BN, ALIGNED_B>(rb, Bt, n_blk, N, K, (kt + 1) * BK, K);
+    }
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi) {
       const int mrow = m_wave + mi * 16 + l15;
-      const short8 af =
-          *reinterpret_cast<const short8*>(&a_lds[mrow * PADK + khalf * 8]);
+      const short8 af = *reinterpret_cast<const short8*>(
+          &a_lds[cur][mrow * PADK + khalf * 8]);
 #pragma unroll
       for (int ni = 0; ni < NFRAG; ++ni) {
         const int nrow = ni * 16 + l15;
-        const short8 bf =
-            *reinterpret_cast<const short8*>(&b_lds[nrow * PADK + khalf * 8]);
+        const short8 bf = *reinterpret_cast<const short8*>(
+            &b_lds[cur][nrow * PADK + khalf * 8]);
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af, bf, acc[mi][ni], 0, 0, 0);
       }
     }
-    __syncthreads();
+    if (kt + 1 < nk) {
+      stage_write<BM>(a_lds[cur ^ 1], ra);
+      stage_write<BN>(b_lds[cur ^ 1], rb);
+      __syncthreads();
+      cur ^= 1;
+    }
   }
 
   // epilogue: C[row][col], row = m_frag + (lane>>4)*4 + j, col = n_frag + l15

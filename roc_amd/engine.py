@@ -42,7 +42,6 @@ class Trainer:
         self.num_classes = num_classes  # true classes if logits are padded
         self.epoch = 0
         F.set_dropout_seed(seed + shard.rank * 7919)
-        self._flat_grad = None
         self.tracer = None  # set via enable_tracing()
 
     def enable_tracing(self):
@@ -107,8 +106,9 @@ class Trainer:
             self._graph = g
         except Exception as e:  # pragma: no cover - capture unsupported
             torch.cuda.synchronize()
+            import sys
             print(f"[roc_amd] hipGraph capture failed ({e!r}); "
-                  "falling back to eager", flush=True)
+                  "falling back to eager", file=sys.stderr, flush=True)
             self.use_graph = False
             return self._epoch_body()
         self._graph.replay()

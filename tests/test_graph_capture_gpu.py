@@ -28,10 +28,15 @@ def _make(graph_capture, dropout=0.3, seed=11):
 
 
 def test_capture_matches_eager_no_dropout():
-    F.set_dropout_counter(None)
-    tr_e = _make(False, dropout=0.0)
+    """Replayed graphs vs the SAME kernels run eagerly. The baseline also
+    uses the device-side Adam schedule (warmup set huge so it never
+    captures) — any difference is capture mechanics alone, bounded by
+    split-K atomic reduction noise."""
+    tr_e = _make(True, dropout=0.0)
+    tr_e._graph_warmup = 10 ** 9  # device-schedule eager forever
     for _ in range(8):
         tr_e.train_epoch()
+    assert tr_e._graph is None
     me = tr_e.evaluate()
     we = tr_e.model.weights[0].detach().cpu()
 
@@ -42,9 +47,7 @@ def test_capture_matches_eager_no_dropout():
     mg = tr_g.evaluate()
     wg = tr_g.model.weights[0].detach().cpu()
     F.set_dropout_counter(None)
-    # identical math modulo split-K atomic ordering + the device-side
-    # __powf Adam schedule (direct kernel equivalence is tested below)
-    assert torch.allclose(we, wg, atol=2e-2), (we - wg).abs().max()
+    assert torch.allclose(we, wg, atol=5e-3), (we - wg).abs().max()
     assert mg["ce_loss"] == pytest.approx(me["ce_loss"], rel=0.05)
     assert tr_g.optimizer.t == tr_e.optimizer.t == 8
 
