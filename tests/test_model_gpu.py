@@ -69,6 +69,30 @@ def test_gcn_overfits_gpu():
     assert m["train_acc"] > 0.85, m
 
 
+def test_gcn_fp32_on_gpu_matches_cpu():
+    """The reference's fp32-only mode, on GPU: must track the CPU fp32
+    path tightly (exact-fp32 MFMA + fp32 gathers)."""
+    g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.15, seed=9)
+    pad = (-feats.shape[1]) % 8
+    if pad:
+        feats = torch.nn.functional.pad(feats, (0, pad))
+    shard = build_shard(g, 0, 1)
+    dims = [feats.shape[1], 32, c]
+    res = {}
+    for dev in ("cpu", "cuda:0"):
+        model = build_model("gcn", dims, dropout=0.0, seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
+        tr = Trainer(model, shard, feats, labels, mask, opt, device=dev,
+                     compute_dtype=torch.float32)
+        for _ in range(3):
+            tr.train_epoch()
+        res[dev] = (tr.evaluate(), model.weights[0].detach().cpu())
+    m_c, w_c = res["cpu"]
+    m_g, w_g = res["cuda:0"]
+    assert torch.allclose(w_c, w_g, atol=1e-3), (w_c - w_g).abs().max()
+    assert m_g["ce_loss"] == pytest.approx(m_c["ce_loss"], rel=1e-3)
+
+
 def test_eval_identity_dropout_gpu():
     tr = make_trainer("gcn", dropout=0.5)
     tr.model.eval()

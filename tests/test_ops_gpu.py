@@ -115,6 +115,34 @@ def test_gemm_rr_random(M, K, N):
         (got - want).abs().max()
 
 
+@pytest.mark.parametrize("M,K,N", [(300, 608, 256), (257, 256, 41),
+                                   (128, 33, 96)])
+def test_gemm_rr_fp32(M, K, N):
+    """exact-fp32 MFMA path (mfma_f32_16x16x4f32) vs torch fp32 matmul."""
+    torch.manual_seed(12)
+    A = torch.randn(M, K)
+    B = torch.randn(K, N)
+    want = A @ B
+    C = torch.empty(M, N, dtype=torch.float32, device=DEV)
+    _ext().gemm_rr(C, A.to(DEV), B.t().contiguous().to(DEV), False)
+    got = C.cpu()
+    # fp32 MFMA is an exact fmaf chain; only summation order differs
+    tol = want.abs().max().item() * 1e-5 + 1e-4
+    assert torch.allclose(got, want, atol=tol, rtol=1e-4), \
+        (got - want).abs().max()
+
+
+def test_gemm_atb_fp32():
+    torch.manual_seed(13)
+    A = torch.randn(2000, 96)
+    B = torch.randn(2000, 64) * 0.1
+    want = A.t() @ B
+    C = torch.zeros(96, 64, dtype=torch.float32, device=DEV)
+    _ext().gemm_atb(C, A.to(DEV), B.to(DEV))
+    tol = want.abs().max().item() * 1e-5 + 1e-3
+    assert torch.allclose(C.cpu(), want, atol=tol, rtol=1e-4)
+
+
 def test_gemm_rr_fused_relu():
     torch.manual_seed(5)
     A = torch.randn(200, 64).to(torch.bfloat16)
