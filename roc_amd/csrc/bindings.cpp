@@ -4,7 +4,7 @@
 void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
           torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,
           c10::optional<torch::Tensor> deg_src,
-          c10::optional<torch::Tensor> row_order);
+          c10::optional<torch::Tensor> row_order, bool accumulate);
 void rowscale(torch::Tensor out, torch::Tensor x, torch::Tensor scale);
 void relu_fwd(torch::Tensor out, torch::Tensor x);
 void sigmoid_fwd(torch::Tensor out, torch::Tensor x);
@@ -28,7 +28,12 @@ void register_graph_cpu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "roc_amd hand-written CDNA4 (gfx950) kernels";
-  m.def("spmm", &spmm, "CSR SpMM neighbor aggregation (fused deg-norm)");
+  m.def("spmm", &spmm, "CSR SpMM neighbor aggregation (fused deg-norm)",
+        pybind11::arg("out"), pybind11::arg("x"), pybind11::arg("rowptr"),
+        pybind11::arg("colidx"), pybind11::arg("deg_dst") = pybind11::none(),
+        pybind11::arg("deg_src") = pybind11::none(),
+        pybind11::arg("row_order") = pybind11::none(),
+        pybind11::arg("accumulate") = false);
   m.def("rowscale", &rowscale);
   m.def("relu_fwd", &relu_fwd);
   m.def("sigmoid_fwd", &sigmoid_fwd);

@@ -129,7 +129,8 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
     T* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
     const float* __restrict__ deg_dst, const float* __restrict__ deg_src,
-    const int* __restrict__ row_order, int num_rows, int64_t D) {
+    const int* __restrict__ row_order, int num_rows, int64_t D,
+    bool accumulate) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const int tpb = kBlock / TEAM;
   const int team = blockIdx.x * tpb + (int)threadIdx.x / TEAM;
@@ -146,27 +147,32 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
       const int64_t e0 = rowptr[row];
       const int64_t e1 = rowptr[row + 1];
       float acc[EPU];
+      T* o = out + (int64_t)row * D + col0;
+      if (accumulate) {  // halo-overlap pass 2: start from the partials
 #pragma unroll
-      for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+        for (int j = 0; j < EPU; ++j) acc[j] = elt_to_f32(o[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+      }
       row_accum_vec<T, EPU, UN8>(acc, x, D, col0, colidx, deg_src, e0, e1);
       if (deg_dst) {
         const float s = deg_dst[row];
 #pragma unroll
         for (int j = 0; j < EPU; ++j) acc[j] *= s;
       }
-      T* o = out + (int64_t)row * D + col0;
       if constexpr (EPU == 8) store_bf16x8(o, acc); else store_f32x4(o, acc);
     }
   } else if (nvalid > 0) {
     for (int ri = team; ri < num_rows; ri += nteams) {
       const int row = row_order ? row_order[ri] : ri;
       float acc[EPU];
-#pragma unroll
-      for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+      T* o = out + (int64_t)row * D + col0;
+      for (int j = 0; j < nvalid; ++j)
+        acc[j] = accumulate ? elt_to_f32(o[j]) : 0.f;
       row_accum_tail<T, EPU>(acc, x, D, col0, nvalid, colidx, deg_src,
                              rowptr[row], rowptr[row + 1]);
       const float s = deg_dst ? deg_dst[row] : 1.f;
-      T* o = out + (int64_t)row * D + col0;
       for (int j = 0; j < nvalid; ++j) f32_to_elt(acc[j] * s, o + j);
     }
   }
@@ -176,7 +182,7 @@ template <typename T>
 void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
                  const float* deg_dst, const float* deg_src,
                  const int* row_order, int num_rows, int64_t D,
-                 hipStream_t stream) {
+                 bool accumulate, hipStream_t stream) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const int64_t units = (D + EPU - 1) / EPU;
   int team = 8;
@@ -194,7 +200,7 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
 #define ROC_SPMM_LAUNCH(TEAM_, UN8_)                                        \
   hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_>), grid, dim3(kBlock), 0,  \
                      stream, out, x, rowptr, colidx, deg_dst, deg_src,      \
-                     row_order, num_rows, D)
+                     row_order, num_rows, D, accumulate)
   switch (team) {
     case 8:
       if (un8) { ROC_SPMM_LAUNCH(8, true); } else { ROC_SPMM_LAUNCH(8, false); }
@@ -216,7 +222,7 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
 void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
           torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,
           c10::optional<torch::Tensor> deg_src,
-          c10::optional<torch::Tensor> row_order) {
+          c10::optional<torch::Tensor> row_order, bool accumulate) {
   ROC_CHECK_DEV_CONT(out);
   ROC_CHECK_DEV_CONT(x);
   ROC_CHECK_DEV_CONT(rowptr);
@@ -239,11 +245,11 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
     launch_spmm<unsigned short>(
         (unsigned short*)out.data_ptr(), (const unsigned short*)x.data_ptr(),
         rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
-        num_rows, D, stream);
+        num_rows, D, accumulate, stream);
   } else if (x.scalar_type() == torch::kFloat32) {
     launch_spmm<float>(out.data_ptr<float>(), x.data_ptr<float>(),
                        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd,
-                       ds, ro, num_rows, D, stream);
+                       ds, ro, num_rows, D, accumulate, stream);
   } else {
     TORCH_CHECK(false, "spmm: unsupported dtype (bf16/f32 only)");
   }

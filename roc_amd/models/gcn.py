@@ -13,7 +13,7 @@ import torch
 
 from ..ops import functional as F
 from ..ops.reference import glorot_uniform
-from ..parallel.halo import halo_exchange
+from ..parallel.halo import halo_exchange, halo_aggregate, overlap_enabled
 
 
 class GCN(torch.nn.Module):
@@ -44,9 +44,15 @@ class GCN(torch.nn.Module):
                 # scales its rows BEFORE the halo exchange); dst-side
                 # D^-1/2 rides the SpMM store. No per-edge degree gather.
                 h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
-                h = halo_exchange(h, shard, group)
-                h = F.scatter_gather(h, shard,
-                                     dst_scale=shard.rsqrt_deg_local)
+                if overlap_enabled() and shard.has_overlap_split:
+                    # interior aggregation overlaps the RCCL exchange
+                    h = halo_aggregate(h, shard,
+                                       dst_scale=shard.rsqrt_deg_local,
+                                       group=group)
+                else:
+                    h = halo_exchange(h, shard, group)
+                    h = F.scatter_gather(h, shard,
+                                         dst_scale=shard.rsqrt_deg_local)
             else:
                 h = F.linear(h, w)
                 h = halo_exchange(h, shard, group)

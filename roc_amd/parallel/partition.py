@@ -51,12 +51,26 @@ class GraphShard:
     recv_splits: List[int]       # rows received from each rank
     send_idx: torch.Tensor       # int64 local indices to send (concat by dest)
     send_splits: List[int]       # rows sent to each rank
+    # --- comm/compute overlap split (world_size > 1 only): local-source
+    # edges aggregate while the halo all_to_all is in flight -------------
+    loc_rowptr: Optional[torch.Tensor] = None   # [n_local+1], cols < n_local
+    loc_colidx: Optional[torch.Tensor] = None
+    halo_rowptr: Optional[torch.Tensor] = None  # [n_local+1], cols < n_halo
+    halo_colidx: Optional[torch.Tensor] = None  # (halo-local ids)
+    t_loc_rowptr: Optional[torch.Tensor] = None   # [n_local+1]
+    t_loc_colidx: Optional[torch.Tensor] = None
+    t_halo_rowptr: Optional[torch.Tensor] = None  # [n_halo+1]
+    t_halo_colidx: Optional[torch.Tensor] = None
 
     def to(self, device) -> "GraphShard":
         d = {}
         for k, v in self.__dict__.items():
             d[k] = v.to(device) if isinstance(v, torch.Tensor) else v
         return GraphShard(**d)
+
+    @property
+    def has_overlap_split(self) -> bool:
+        return self.loc_rowptr is not None
 
 
 def edge_balanced_bounds(rowptr: torch.Tensor, num_parts: int) -> List[int]:
@@ -155,11 +169,35 @@ def build_shard(g: CSRGraph, rank: int, world_size: int,
     t_deg = np.diff(t_rowptr.numpy())
     t_row_order = np.argsort(-t_deg, kind="stable").astype(np.int32)
 
+    shard_kw = {}
+    if world_size > 1:
+        # split edges by source locality for comm/compute overlap
+        is_loc_edge = colidx < n_local
+        row_of_edge = np.repeat(np.arange(n_local, dtype=np.int64),
+                                np.diff(local_rowptr))
+        for key_pfx, sel in (("loc", is_loc_edge), ("halo", ~is_loc_edge)):
+            rows_sel = row_of_edge[sel]
+            cols_sel = colidx[sel].astype(np.int32)
+            if key_pfx == "halo":
+                cols_sel = cols_sel - n_local  # halo-local id space
+            cnt = np.bincount(rows_sel, minlength=n_local).astype(np.int64)
+            rp = np.zeros(n_local + 1, dtype=np.int64)
+            np.cumsum(cnt, out=rp[1:])
+            shard_kw[f"{key_pfx}_rowptr"] = torch.from_numpy(rp)
+            shard_kw[f"{key_pfx}_colidx"] = torch.from_numpy(
+                np.ascontiguousarray(cols_sel))
+        t_loc = build_transpose(n_local, shard_kw["loc_rowptr"],
+                                shard_kw["loc_colidx"])
+        t_halo = build_transpose(max(n_halo, 0), shard_kw["halo_rowptr"],
+                                 shard_kw["halo_colidx"])
+        shard_kw["t_loc_rowptr"], shard_kw["t_loc_colidx"] = t_loc
+        shard_kw["t_halo_rowptr"], shard_kw["t_halo_colidx"] = t_halo
+
     return GraphShard(
         rank=rank, world_size=world_size, bounds=list(bounds),
         lo=lo, hi=hi, n_local=n_local, n_halo=n_halo, n_ext=n_ext,
         rowptr=rowptr_t, colidx=colidx_t,
-        t_rowptr=t_rowptr, t_colidx=t_colidx,
+        t_rowptr=t_rowptr, t_colidx=t_colidx, **shard_kw,
         deg_local=torch.from_numpy(deg_local.copy()),
         rsqrt_deg_local=torch.from_numpy(1.0 / np.sqrt(deg_local)),
         inv_deg_local=torch.from_numpy(1.0 / deg_local),
