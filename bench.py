@@ -103,7 +103,11 @@ def main():
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
                       compute_dtype=dtype, grad_scale=1.0, seed=args.seed,
                       num_classes=num_classes)
-    if device != "cpu" and not args.no_graph:
+    # hipGraph capture is default-on for 1 GPU; for multi-GPU it wraps
+    # RCCL collectives in the graph — enable explicitly once measured
+    # (ROC_GRAPH_MULTI=1). Capture failure falls back to eager either way.
+    if device != "cpu" and not args.no_graph and (
+            world == 1 or os.environ.get("ROC_GRAPH_MULTI") == "1"):
         trainer.enable_graph_capture()
 
     def barrier():

@@ -85,3 +85,30 @@ def test_train_cli_runs(tmp_path):
         timeout=300)
     assert r2.returncode == 0, r2.stderr[-2000:]
     assert "resumed" in r2.stdout
+
+
+def test_auto_recover_cli(tmp_path):
+    """Divergence guard + checkpoint restore: train with an absurd LR that
+    NaNs out; --auto-recover must restore and finish."""
+    import numpy as np
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ck = str(tmp_path / "r.pt")
+    # first produce a good checkpoint
+    r = subprocess.run(
+        [sys.executable, "train.py", "--dataset", "cora-synthetic",
+         "--scale", "0.04", "--epochs", "3", "--hidden", "8",
+         "--eval-every", "0", "--checkpoint", ck],
+        cwd=env["PYTHONPATH"], env=env, capture_output=True, text=True,
+        timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    # resume with a huge LR: must diverge, recover, and complete
+    r2 = subprocess.run(
+        [sys.executable, "train.py", "--dataset", "cora-synthetic",
+         "--scale", "0.04", "--epochs", "8", "--hidden", "8",
+         "--eval-every", "1", "--resume", ck, "--checkpoint", ck,
+         "--auto-recover", "--lr", "1e18"],
+        cwd=env["PYTHONPATH"], env=env, capture_output=True, text=True,
+        timeout=300)
+    assert r2.returncode == 0, (r2.stdout[-800:], r2.stderr[-800:])
+    assert "[recover]" in r2.stdout, r2.stdout[-800:]

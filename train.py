@@ -26,6 +26,7 @@ from roc_amd.graph import (load_lux, load_features, load_labels, load_mask,
                            DATASET_SHAPES)
 from roc_amd.parallel.partition import edge_balanced_bounds
 from roc_amd.utils import save_checkpoint, load_checkpoint
+from roc_amd.debug import check_metrics, TrainingDiverged
 
 
 def parse_args():
@@ -64,6 +65,12 @@ def parse_args():
                     help="write a chrome trace JSON here at the end")
     ap.add_argument("--offload", action="store_true",
                     help="host-DRAM activation offload (capacity tier)")
+    ap.add_argument("--rebalance-every", type=int, default=0,
+                    help="cost-model repartition every N epochs (measured "
+                         "per-rank throughput; multi-rank only)")
+    ap.add_argument("--auto-recover", action="store_true",
+                    help="on divergence (non-finite loss): restore the "
+                         "last checkpoint, halve the LR, continue")
     ap.add_argument("-v", "--verbose", action="store_true")
     args = ap.parse_args()
     if args.config:
@@ -164,10 +171,32 @@ def main():
               f"dtype={dtype} lr={args.lr} wd={args.weight_decay} "
               f"dropout={args.dropout}", flush=True)
 
+    if args.rebalance_every:
+        trainer.attach_full_graph(g)
+
     t_start = time.perf_counter()
     while trainer.epoch < args.epochs:
-        trainer.train_epoch()
+        metrics = trainer.train_epoch()
         ep = trainer.epoch
+        if args.auto_recover and (args.eval_every == 0
+                                  or ep % max(args.eval_every, 1) == 0):
+            try:
+                check_metrics(metrics)
+            except TrainingDiverged as e:
+                if not (args.checkpoint and os.path.exists(args.checkpoint)):
+                    raise
+                if rank == 0:
+                    print(f"[recover] {e}; restoring {args.checkpoint}, "
+                          f"lr {trainer.optimizer.lr} -> "
+                          f"{trainer.optimizer.lr * 0.5}", flush=True)
+                load_checkpoint(args.checkpoint, trainer)
+                trainer.optimizer.lr *= 0.5
+                continue
+        if (args.rebalance_every and ep % args.rebalance_every == 0
+                and world > 1):
+            nb = trainer.measure_and_rebalance(feats, labels, mask)
+            if rank == 0 and args.verbose:
+                print(f"[rebalance] epoch {ep}: bounds -> {nb}", flush=True)
         if args.eval_every and ep % args.eval_every == 0:
             md = trainer.evaluate()
             if rank == 0:
