@@ -53,6 +53,15 @@ def main():
         out = torch.empty_like(x)
         gb = args.edges * D * 2 / 1e9
         print(f"== D={D}  gather volume {gb:.1f} GB ==")
+        # locality ceiling probe: confine sources to an L2-scale window
+        for window in (4096, 16384, 65536):
+            ci_w = torch.remainder(colidx, window)
+            med, best = time_variant(
+                lambda: _C.spmm(out, x, rowptr, ci_w, rsq, None, None),
+                args.rounds)
+            mb = window * D * 2 / 1e6
+            print(f"  src-window {window:6d} rows ({mb:6.1f} MB): "
+                  f"median {med:8.2f} ms ({gb/med:.2f} TB/s eff)", flush=True)
         variants = {}
         for un8 in (0, 1):
             for use_order in (0, 1):
