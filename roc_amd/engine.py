@@ -27,14 +27,19 @@ class Trainer:
                  optimizer: AdamOptimizer, device="cpu",
                  compute_dtype: torch.dtype = torch.float32,
                  grad_scale: float = 1.0, group=None, seed: int = 1,
-                 num_classes=None):
+                 num_classes=None, local_slices: bool = False):
+        # local_slices=True: feats/labels/mask are ALREADY this rank's
+        # [lo, hi) rows (windowed dataset loading)
         self.model = model.to(device)
         self.shard = shard.to(device)
         self.device = torch.device(device)
         self.dtype = compute_dtype
         self.group = group
         self.grad_scale = grad_scale
-        lo, hi = shard.lo, shard.hi
+        if local_slices:
+            lo, hi = 0, shard.n_local
+        else:
+            lo, hi = shard.lo, shard.hi
         self.x = feats[lo:hi].to(device=device, dtype=compute_dtype).contiguous()
         self.labels = labels[lo:hi].to(device=device).contiguous()
         self.mask = mask[lo:hi].to(device=device, dtype=torch.int32).contiguous()
@@ -200,9 +205,11 @@ class Trainer:
         t = self.timed_epochs(probe_epochs) / probe_epochs
         if self.shard.world_size == 1 or not dist.is_initialized():
             return self.shard.bounds
-        times = torch.zeros(self.shard.world_size)
+        dev = self.device if self.device.type == "cuda" else "cpu"
+        times = torch.zeros(self.shard.world_size, device=dev)
         times[self.shard.rank] = t
         dist.all_reduce(times, group=self.group)
+        times = times.cpu()
         # rowptr of the full graph is not kept; derive per-rank edge counts
         # from the current shard (identical on all ranks via bounds)
         new_bounds = rebalance_bounds(

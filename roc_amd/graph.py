@@ -150,6 +150,41 @@ def load_lux(path: str) -> CSRGraph:
     )
 
 
+def load_lux_meta(path: str):
+    """Header + full row-offset array only (numNodes*8 bytes — the part
+    every rank needs for partitioning/degrees; the edge list is read
+    per-partition with load_lux_cols, reference `load_task.cu:231-243`)."""
+    with open(path, "rb") as f:
+        num_nodes = struct.unpack("<I", f.read(4))[0]
+        num_edges = struct.unpack("<Q", f.read(8))[0]
+        raw_rows = np.fromfile(f, dtype=np.uint64, count=num_nodes)
+    rowptr = np.zeros(num_nodes + 1, dtype=np.int64)
+    rowptr[1:] = raw_rows.astype(np.int64)
+    return num_nodes, num_edges, torch.from_numpy(rowptr)
+
+
+def load_lux_cols(path: str, num_nodes: int, e0: int, e1: int) -> np.ndarray:
+    """Windowed read of source ids for edges [e0, e1)."""
+    header = 4 + 8 + 8 * num_nodes
+    with open(path, "rb") as f:
+        f.seek(header + 4 * e0)
+        cols = np.fromfile(f, dtype=np.uint32, count=e1 - e0)
+    return cols.astype(np.int64)
+
+
+def load_features_window(path_prefix: str, num_nodes: int, in_dim: int,
+                         lo: int, hi: int) -> torch.Tensor:
+    """Rows [lo, hi) of the feature matrix from the binary cache
+    (falls back to loading+caching the CSV once)."""
+    bin_path = path_prefix + ".feats.bin"
+    if not os.path.exists(bin_path):
+        load_features(path_prefix, num_nodes, in_dim)  # writes the cache
+    with open(bin_path, "rb") as f:
+        f.seek(4 * in_dim * lo)
+        arr = np.fromfile(f, dtype=np.float32, count=(hi - lo) * in_dim)
+    return torch.from_numpy(arr.reshape(hi - lo, in_dim).copy())
+
+
 def save_lux(path: str, g: CSRGraph) -> None:
     with open(path, "wb") as f:
         f.write(struct.pack("<I", g.num_nodes))

@@ -107,18 +107,70 @@ def rebalance_bounds(rowptr: torch.Tensor, bounds: List[int],
     return nb
 
 
+def _send_plan_comm(halo_ids: np.ndarray, recv_splits: List[int], lo: int,
+                    world_size: int, group):
+    """Exchange halo REQUESTS over the process group: each rank learns
+    which of its rows the others need without scanning their edge
+    windows (required for windowed dataset loading at scale)."""
+    import torch.distributed as dist
+    counts_in = torch.tensor(recv_splits, dtype=torch.int64)
+    counts_out = torch.empty(world_size, dtype=torch.int64)
+    dist.all_to_all_single(counts_out, counts_in, group=group)
+    send_splits = [int(c) for c in counts_out]
+    req = torch.empty(sum(send_splits), dtype=torch.int64)
+    dist.all_to_all_single(req, torch.from_numpy(halo_ids.astype(np.int64)),
+                           output_split_sizes=send_splits,
+                           input_split_sizes=recv_splits, group=group)
+    return (req - lo), send_splits
+
+
 def build_shard(g: CSRGraph, rank: int, world_size: int,
-                bounds: Optional[List[int]] = None) -> GraphShard:
-    """Build this rank's shard from the full graph (all ranks deterministic)."""
+                bounds: Optional[List[int]] = None,
+                use_comm: bool = False, group=None) -> GraphShard:
+    """Build this rank's shard from the full graph. The halo SEND plan
+    comes either from scanning every rank's edge window (deterministic,
+    no process group needed) or — with use_comm=True under an initialized
+    torch.distributed group — from an all_to_all of halo requests (each
+    rank touches only its own window; scales to windowed file loading)."""
     if bounds is None:
         bounds = edge_balanced_bounds(g.rowptr, world_size)
     lo, hi = bounds[rank], bounds[rank + 1]
-    n_local = hi - lo
     rp = g.rowptr.numpy()
-    ci = g.colidx.numpy()
     e0, e1 = int(rp[lo]), int(rp[hi])
+    local_cols_global = g.colidx.numpy()[e0:e1].astype(np.int64)
+    return build_shard_from_window(
+        g.rowptr, local_cols_global, rank, world_size, bounds,
+        full_colidx=None if use_comm else g.colidx, group=group)
+
+
+def build_shard_from_lux(path: str, rank: int, world_size: int,
+                         bounds: Optional[List[int]] = None,
+                         group=None) -> GraphShard:
+    """Windowed shard construction straight from a .lux file: this rank
+    reads only its own edge window (reference `load_task.cu:231-243`);
+    the send plan rides the process group (world_size > 1)."""
+    from ..graph import load_lux_meta, load_lux_cols
+    num_nodes, num_edges, rowptr = load_lux_meta(path)
+    if bounds is None:
+        bounds = edge_balanced_bounds(rowptr, world_size)
+    lo, hi = bounds[rank], bounds[rank + 1]
+    rp = rowptr.numpy()
+    cols = load_lux_cols(path, num_nodes, int(rp[lo]), int(rp[hi]))
+    return build_shard_from_window(rowptr, cols, rank, world_size, bounds,
+                                   full_colidx=None, group=group)
+
+
+def build_shard_from_window(rowptr_full: torch.Tensor,
+                            local_cols_global: np.ndarray, rank: int,
+                            world_size: int, bounds: List[int],
+                            full_colidx: Optional[torch.Tensor] = None,
+                            group=None) -> GraphShard:
+    lo, hi = bounds[rank], bounds[rank + 1]
+    n_local = hi - lo
+    rp = rowptr_full.numpy()
+    e0 = int(rp[lo])
     local_rowptr = (rp[lo:hi + 1] - e0).astype(np.int64)
-    local_cols_global = ci[e0:e1].astype(np.int64)
+    local_cols_global = local_cols_global.astype(np.int64, copy=False)
 
     # halo: remote source vertices, grouped by owning rank then sorted by id
     is_local = (local_cols_global >= lo) & (local_cols_global < hi)
@@ -141,22 +193,30 @@ def build_shard(g: CSRGraph, rank: int, world_size: int,
     colidx_t = torch.from_numpy(colidx)
     t_rowptr, t_colidx = build_transpose(n_ext, rowptr_t, colidx_t)
 
-    # who needs MY rows: every rank computes every rank's halo deterministically
-    send_chunks, send_splits = [], []
-    for r in range(world_size):
-        if r == rank:
-            send_chunks.append(np.empty(0, dtype=np.int64))
-            send_splits.append(0)
-            continue
-        rlo, rhi = bounds[r], bounds[r + 1]
-        their_cols = ci[rp[rlo]:rp[rhi]]
-        theirs_from_me = np.unique(
-            their_cols[(their_cols >= lo) & (their_cols < hi)]
-        ).astype(np.int64)
-        send_chunks.append(theirs_from_me - lo)
-        send_splits.append(theirs_from_me.shape[0])
-    send_idx = torch.from_numpy(np.concatenate(send_chunks)) if world_size > 1 \
-        else torch.empty(0, dtype=torch.int64)
+    # who needs MY rows
+    if world_size == 1:
+        send_idx = torch.empty(0, dtype=torch.int64)
+        send_splits = [0]
+    elif full_colidx is not None:
+        # scan every rank's window (no process group needed; deterministic)
+        ci = full_colidx.numpy()
+        send_chunks, send_splits = [], []
+        for r in range(world_size):
+            if r == rank:
+                send_chunks.append(np.empty(0, dtype=np.int64))
+                send_splits.append(0)
+                continue
+            rlo, rhi = bounds[r], bounds[r + 1]
+            their_cols = ci[rp[rlo]:rp[rhi]]
+            theirs_from_me = np.unique(
+                their_cols[(their_cols >= lo) & (their_cols < hi)]
+            ).astype(np.int64)
+            send_chunks.append(theirs_from_me - lo)
+            send_splits.append(theirs_from_me.shape[0])
+        send_idx = torch.from_numpy(np.concatenate(send_chunks))
+    else:
+        send_idx, send_splits = _send_plan_comm(
+            halo_ids, recv_splits, lo, world_size, group)
 
     # degrees (GLOBAL in-degree — normalization must match the 1-GPU model)
     deg_all = np.maximum(np.diff(rp), 1).astype(np.float32)

@@ -92,19 +92,33 @@ def parse_args():
     return args
 
 
-def load_dataset(args):
+def load_dataset(args, rank=0, world=1):
+    """Synthetic shapes, or reference-format files. With --file and
+    world > 1, each rank reads ONLY its window of the edge list and
+    feature matrix (reference `load_task.cu:231-243` parity); the halo
+    send plan is exchanged over the process group."""
     if args.file:
-        g = load_lux(args.file + ".add_self_edge.lux")
-        # infer in_dim from the layers flag (reference requires -layers)
         assert args.layers, "--file requires --layers D0-...-C"
         dims = [int(d) for d in args.layers.split("-")]
+        lux = args.file + ".add_self_edge.lux"
+        if world > 1:
+            from roc_amd.parallel.partition import build_shard_from_lux
+            from roc_amd.graph import load_features_window, load_lux_meta
+            shard = build_shard_from_lux(lux, rank, world)
+            n = load_lux_meta(lux)[0]
+            feats = load_features_window(args.file, n, dims[0],
+                                         shard.lo, shard.hi)
+            labels = load_labels(args.file + ".label", n)[shard.lo:shard.hi]
+            mask = load_mask(args.file + ".mask", n)[shard.lo:shard.hi]
+            return (None, feats, labels, mask, dims[-1]), shard
+        g = load_lux(lux)
         feats = load_features(args.file, g.num_nodes, dims[0])
         labels = load_labels(args.file + ".label", g.num_nodes)
         mask = load_mask(args.file + ".mask", g.num_nodes)
-        return g, feats, labels, mask, dims[-1]
+        return (g, feats, labels, mask, dims[-1]), None
     name = args.dataset.replace("-synthetic", "")
     assert name in DATASET_SHAPES, f"unknown dataset {name}"
-    return synthetic_dataset(name, seed=args.seed, scale=args.scale)
+    return synthetic_dataset(name, seed=args.seed, scale=args.scale), None
 
 
 def main():
@@ -122,7 +136,8 @@ def main():
     else:
         device = "cuda:0" if on_gpu else "cpu"
 
-    g, feats, labels, mask, num_classes = load_dataset(args)
+    (g, feats, labels, mask, num_classes), pre_shard = \
+        load_dataset(args, rank, world)
     # pad feature dim for 16-B-aligned bf16 rows (zero cols; exact math)
     pad = (-feats.shape[1]) % 8
     if pad and on_gpu:
@@ -138,8 +153,12 @@ def main():
         dims = [feats.shape[1]] + [args.hidden] * (args.num_layers - 1) \
             + [c_out]
 
-    bounds = edge_balanced_bounds(g.rowptr, world)
-    shard = build_shard(g, rank, world, bounds)
+    if pre_shard is not None:
+        shard = pre_shard
+    else:
+        bounds = edge_balanced_bounds(g.rowptr, world)
+        shard = build_shard(g, rank, world, bounds,
+                            use_comm=(world > 1 and dist.is_initialized()))
     mkw = {"residual": args.residual} if args.model == "gcn" else {}
     model = build_model(args.model, dims, dropout=args.dropout,
                         seed=args.seed, **mkw)
@@ -155,7 +174,7 @@ def main():
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
                       compute_dtype=dtype, grad_scale=gs, seed=args.seed,
                       num_classes=num_classes if dims[-1] != num_classes
-                      else None)
+                      else None, local_slices=pre_shard is not None)
     if args.trace:
         trainer.enable_tracing()
     if args.offload:
