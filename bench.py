@@ -135,8 +135,13 @@ def main():
 
     per_epoch = elapsed / args.steps
     if rank == 0:
+        metric_name = ("per-epoch training time (s), 2-layer GCN on Reddit"
+                       if (args.model, args.layers, args.dataset)
+                       == ("gcn", 2, "reddit")
+                       else f"per-epoch training time (s), {args.layers}-layer "
+                            f"{args.model} on {args.dataset}")
         result = {
-            "metric": "per-epoch training time (s), 2-layer GCN on Reddit",
+            "metric": metric_name,
             "value": per_epoch,
             "unit": "s/epoch",
             "n_gpus": world,

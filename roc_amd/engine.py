@@ -82,6 +82,8 @@ class Trainer:
         F.reset_dropout_offset()
         self.optimizer.zero_grad()
         loss, metrics = self._forward_loss()
+        if self.offload is not None:
+            self.offload.prefetch()
         loss.backward()
         self._allreduce_grads()
         self.optimizer.step()
@@ -159,6 +161,8 @@ class Trainer:
         if self.tracer is None:
             self.optimizer.zero_grad()
             loss, metrics = self._forward_loss()
+            if self.offload is not None:
+                self.offload.prefetch()
             loss.backward()
             self._allreduce_grads()
             self.optimizer.step()
@@ -169,6 +173,8 @@ class Trainer:
             with tr.span("forward"):
                 loss, metrics = self._forward_loss()
             with tr.span("backward"):
+                if self.offload is not None:
+                    self.offload.prefetch()
                 loss.backward()
             with tr.span("grad_allreduce"):
                 self._allreduce_grads()

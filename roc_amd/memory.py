@@ -25,13 +25,16 @@ class ActivationOffload:
         loss = model(...); loss.backward()
     """
 
-    def __init__(self, min_bytes: int = 1 << 22, enabled: bool = True):
+    def __init__(self, min_bytes: int = 1 << 22, enabled: bool = True,
+                 prefetch: bool = True):
         self.min_bytes = min_bytes
         self.enabled = enabled and torch.cuda.is_available()
+        self.prefetch = prefetch
         self.stream = torch.cuda.Stream() if self.enabled else None
         self._pool = {}   # (shape, dtype) -> list of free pinned tensors
         self._stats = {"offloaded_bytes": 0, "tensors": 0}
         self._ctx = None
+        self._entries = []  # live offloaded entries, forward order
 
     # -- pinned buffer pool -------------------------------------------------
     def _acquire(self, t: torch.Tensor) -> torch.Tensor:
@@ -62,22 +65,64 @@ class ActivationOffload:
         t.record_stream(self.stream)          # allocator: defer reuse
         self._stats["offloaded_bytes"] += t.numel() * t.element_size()
         self._stats["tensors"] += 1
-        return ("roc_offloaded", cpu, t.device, ev)
+        entry = {"cpu": cpu, "device": t.device, "ev": ev,
+                 "gpu": None, "ev2": None}
+        self._entries.append(entry)
+        return ("roc_offloaded", entry)
+
+    def prefetch(self, window: int = 4):
+        """Start H2D copies for the next `window` offloaded tensors in
+        LIFO order (backward consumes saved tensors roughly in reverse
+        forward order); each consumption triggers the next copy. Keeps at
+        most `window` prefetched activations resident — capacity stays
+        bounded. Call right before loss.backward()."""
+        if not self.enabled:
+            return
+        self._order = list(reversed(self._entries))
+        self._pf_ptr = 0
+        for _ in range(min(window, len(self._order))):
+            self._start_next()
+
+    _order = ()
+    _pf_ptr = 0
+
+    def _start_next(self):
+        while self._pf_ptr < len(self._order):
+            e = self._order[self._pf_ptr]
+            self._pf_ptr += 1
+            if e["gpu"] is None and e["cpu"] is not None:
+                self._start_h2d(e)
+                return
+
+    def _start_h2d(self, entry):
+        with torch.cuda.stream(self.stream):
+            entry["ev"].wait(self.stream)     # D2H done before H2D
+            entry["gpu"] = entry["cpu"].to(entry["device"], non_blocking=True)
+            ev2 = torch.cuda.Event()
+            ev2.record(self.stream)
+            entry["ev2"] = ev2
 
     def _unpack(self, packed):
         if not isinstance(packed, tuple) or not packed or \
                 packed[0] != "roc_offloaded":
             return packed
-        _, cpu, device, ev = packed
-        cur = torch.cuda.current_stream(device)
-        with torch.cuda.stream(self.stream):
-            ev.wait(self.stream)              # D2H done before reuse/H2D
-            gpu = cpu.to(device, non_blocking=True)
-            ev2 = torch.cuda.Event()
-            ev2.record(self.stream)
-        cur.wait_event(ev2)
+        entry = packed[1]
+        if entry["cpu"] is None:
+            raise RuntimeError(
+                "offloaded activation already consumed (double backward is "
+                "not supported with ActivationOffload)")
+        if entry["gpu"] is None:
+            self._start_h2d(entry)
+        cur = torch.cuda.current_stream(entry["device"])
+        cur.wait_event(entry["ev2"])
+        gpu = entry["gpu"]
         gpu.record_stream(cur)
-        self._release(cpu)
+        self._release(entry["cpu"])
+        entry["cpu"] = None
+        entry["gpu"] = None
+        if entry in self._entries:
+            self._entries.remove(entry)
+        self._start_next()                    # keep the pipeline full
         return gpu
 
     def __enter__(self):
