@@ -197,10 +197,12 @@ class _Linear(torch.autograd.Function):
                 dy = dym
             dw = torch.zeros_like(w)  # fp32 [in, out]
             _C.gemm_atb(dw, x, dy)
-            # dx = dy @ w^T: gemm_rr's Bt = (w^T)^T = w as stored [in,out]
-            wc = w.contiguous().to(dy.dtype)
-            dx = torch.empty_like(x)
-            _C.gemm_rr(dx, dy, wc, False)
+            dx = None
+            if ctx.needs_input_grad[0]:  # layer-1 inputs carry no grad
+                # dx = dy @ w^T: gemm_rr's Bt = (w^T)^T = w [in,out]
+                wc = w.contiguous().to(dy.dtype)
+                dx = torch.empty_like(x)
+                _C.gemm_rr(dx, dy, wc, False)
         else:
             if act == "relu":
                 dy = ref.relu_grad(dy, y)
@@ -209,7 +211,7 @@ class _Linear(torch.autograd.Function):
             if row_scale is not None:
                 dy = dy * row_scale.unsqueeze(1).to(dy.dtype)
             dw = (x.to(torch.float32).t() @ dy.to(torch.float32))
-            dx = (dy @ w.t().to(dy.dtype))
+            dx = (dy @ w.t().to(dy.dtype)) if ctx.needs_input_grad[0] else None
         return dx, dw, None, None
 
 
