@@ -26,10 +26,10 @@ class ActivationOffload:
     """
 
     def __init__(self, min_bytes: int = 1 << 22, enabled: bool = True,
-                 prefetch: bool = True):
+                 prefetch_window: int = 4):
         self.min_bytes = min_bytes
         self.enabled = enabled and torch.cuda.is_available()
-        self.prefetch = prefetch
+        self.prefetch_window = prefetch_window
         self.stream = torch.cuda.Stream() if self.enabled else None
         self._pool = {}   # (shape, dtype) -> list of free pinned tensors
         self._stats = {"offloaded_bytes": 0, "tensors": 0}
@@ -70,7 +70,7 @@ class ActivationOffload:
         self._entries.append(entry)
         return ("roc_offloaded", entry)
 
-    def prefetch(self, window: int = 4):
+    def prefetch(self, window: int = 0):
         """Start H2D copies for the next `window` offloaded tensors in
         LIFO order (backward consumes saved tensors roughly in reverse
         forward order); each consumption triggers the next copy. Keeps at
@@ -78,6 +78,7 @@ class ActivationOffload:
         bounded. Call right before loss.backward()."""
         if not self.enabled:
             return
+        window = window or self.prefetch_window
         self._order = list(reversed(self._entries))
         self._pf_ptr = 0
         for _ in range(min(window, len(self._order))):
