@@ -27,8 +27,11 @@ class ActivationOffload:
 
     def __init__(self, min_bytes: int = 1 << 22, enabled: bool = True,
                  prefetch_window: int = 4):
+        import os
         self.min_bytes = min_bytes
         self.enabled = enabled and torch.cuda.is_available()
+        if os.environ.get("ROC_OFFLOAD_PREFETCH", "1") == "0":
+            prefetch_window = 0
         self.prefetch_window = prefetch_window
         self.stream = torch.cuda.Stream() if self.enabled else None
         self._pool = {}   # (shape, dtype) -> list of free pinned tensors
@@ -52,6 +55,7 @@ class ActivationOffload:
     # -- hooks ---------------------------------------------------------------
     def _pack(self, t: torch.Tensor):
         if (not self.enabled or not t.is_cuda
+                or not t.is_floating_point()  # CSR indices etc stay put
                 or t.numel() * t.element_size() < self.min_bytes
                 or (t.is_leaf and t.requires_grad)):  # keep params resident
             return t
@@ -76,7 +80,7 @@ class ActivationOffload:
         forward order); each consumption triggers the next copy. Keeps at
         most `window` prefetched activations resident — capacity stays
         bounded. Call right before loss.backward()."""
-        if not self.enabled:
+        if not self.enabled or self.prefetch_window <= 0:
             return
         window = window or self.prefetch_window
         self._order = list(reversed(self._entries))

@@ -61,6 +61,13 @@ class GraphShard:
     t_loc_colidx: Optional[torch.Tensor] = None
     t_halo_rowptr: Optional[torch.Tensor] = None  # [n_halo+1]
     t_halo_colidx: Optional[torch.Tensor] = None
+    # --- exchange strategy (parallel/aggregate.py) ----------------------
+    comm_mode: str = "none"        # none | halo | allgather
+    halo_fraction: float = 0.0     # n_halo / remote rows
+    ag_max_rows: int = 0           # padded block size for all_gather
+    ag_colidx: Optional[torch.Tensor] = None     # cols in gather space
+    ag_t_rowptr: Optional[torch.Tensor] = None   # [ws*ag_max_rows+1]
+    ag_t_colidx: Optional[torch.Tensor] = None
 
     def to(self, device) -> "GraphShard":
         d = {}
@@ -231,6 +238,28 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
 
     shard_kw = {}
     if world_size > 1:
+        num_nodes_total = bounds[-1]
+        halo_fraction = n_halo / max(num_nodes_total - n_local, 1)
+        import os as _os
+        mode = _os.environ.get("ROC_COMM_MODE", "auto")
+        if mode == "auto":
+            mode = "allgather" if halo_fraction > 0.5 else "halo"
+        shard_kw["comm_mode"] = mode
+        shard_kw["halo_fraction"] = float(halo_fraction)
+    if world_size > 1 and shard_kw.get("comm_mode") == "allgather":
+        # gather-space CSR: node (owner r, local i) -> r*max_rows + i
+        sizes = [bounds[r + 1] - bounds[r] for r in range(world_size)]
+        mr = max(sizes)
+        owners = np.searchsorted(bounds, local_cols_global,
+                                 side="right") - 1
+        ag_cols = (owners.astype(np.int64) * mr +
+                   (local_cols_global - np.asarray(bounds)[owners]))
+        shard_kw["ag_max_rows"] = int(mr)
+        shard_kw["ag_colidx"] = torch.from_numpy(ag_cols.astype(np.int32))
+        ag_t = build_transpose(world_size * mr, rowptr_t,
+                               shard_kw["ag_colidx"])
+        shard_kw["ag_t_rowptr"], shard_kw["ag_t_colidx"] = ag_t
+    if world_size > 1 and shard_kw.get("comm_mode") == "halo":
         # split edges by source locality for comm/compute overlap
         is_loc_edge = colidx < n_local
         row_of_edge = np.repeat(np.arange(n_local, dtype=np.int64),
