@@ -13,7 +13,8 @@ import torch
 
 from ..ops import functional as F
 from ..ops.reference import glorot_uniform
-from ..parallel.halo import halo_exchange, halo_aggregate, overlap_enabled
+from ..parallel.halo import halo_exchange
+from ..parallel.aggregate import aggregate
 
 
 class GCN(torch.nn.Module):
@@ -44,15 +45,10 @@ class GCN(torch.nn.Module):
                 # scales its rows BEFORE the halo exchange); dst-side
                 # D^-1/2 rides the SpMM store. No per-edge degree gather.
                 h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
-                if overlap_enabled() and shard.has_overlap_split:
-                    # interior aggregation overlaps the RCCL exchange
-                    h = halo_aggregate(h, shard,
-                                       dst_scale=shard.rsqrt_deg_local,
-                                       group=group)
-                else:
-                    h = halo_exchange(h, shard, group)
-                    h = F.scatter_gather(h, shard,
-                                         dst_scale=shard.rsqrt_deg_local)
+                # strategy (halo a2a / overlap / all_gather) is the
+                # shard's choice — see parallel/aggregate.py
+                h = aggregate(h, shard, dst_scale=shard.rsqrt_deg_local,
+                              group=group)
             else:
                 h = F.linear(h, w)
                 h = halo_exchange(h, shard, group)

@@ -9,7 +9,7 @@ import torch
 
 from ..ops import functional as F
 from ..ops.reference import glorot_uniform
-from ..parallel.halo import halo_exchange, halo_aggregate, overlap_enabled
+from ..parallel.aggregate import aggregate
 
 
 class GIN(torch.nn.Module):
@@ -33,11 +33,7 @@ class GIN(torch.nn.Module):
         nlayers = len(self.w1)
         for i in range(nlayers):
             h = F.dropout(x, self.p, self.training)
-            if overlap_enabled() and shard.has_overlap_split:
-                agg = halo_aggregate(h, shard, group=group)
-            else:
-                hx = halo_exchange(h, shard, group)
-                agg = F.scatter_gather(hx, shard)
+            agg = aggregate(h, shard, group=group)
             h = agg + (1.0 + self.eps[i]).to(h.dtype) * h  # scalar-eps glue
             h = F.linear(h, self.w1[i], activation="relu")
             h = F.linear(h, self.w2[i])

@@ -13,7 +13,7 @@ import torch
 
 from ..ops import functional as F
 from ..ops.reference import glorot_uniform
-from ..parallel.halo import halo_exchange, halo_aggregate, overlap_enabled
+from ..parallel.aggregate import aggregate
 
 
 class GraphSAGE(torch.nn.Module):
@@ -35,14 +35,8 @@ class GraphSAGE(torch.nn.Module):
             h = F.dropout(x, self.p, self.training)
             h_self = F.linear(h, self.w_self[i])
             hn = F.linear(h, self.w_neigh[i])
-            if overlap_enabled() and shard.has_overlap_split:
-                hn = halo_aggregate(hn, shard,
-                                    dst_scale=shard.inv_deg_local,
-                                    group=group)  # fused mean
-            else:
-                hn = halo_exchange(hn, shard, group)
-                hn = F.scatter_gather(hn, shard,
-                                      dst_scale=shard.inv_deg_local)  # mean
+            hn = aggregate(hn, shard, dst_scale=shard.inv_deg_local,
+                           group=group)  # fused mean
             h = F.add(h_self, hn)
             if i < nlayers - 1:
                 h = F.relu(h)

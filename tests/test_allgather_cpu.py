@@ -1,57 +1,72 @@
-"""Comm/compute-overlap halo aggregation == sequential halo+aggregate
-(gloo, world_size=2, all three model families)."""
+"""All-gather/reduce-scatter aggregation == halo-exchange aggregation
+(gloo, world_size=2). Also checks the auto strategy pick."""
 import os
 
-import numpy as np
 import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
-from roc_amd.graph import synthetic_dataset
+from roc_amd.graph import synthetic_dataset, synthetic_graph
 from roc_amd.parallel.partition import build_shard, edge_balanced_bounds
 from roc_amd import build_model, AdamOptimizer, Trainer
 
 WS = 2
 
 
+def test_auto_mode_picks_allgather_for_uniform():
+    g = synthetic_graph(400, 20000, seed=3)  # uniform -> near-total halo
+    sh = build_shard(g, 0, 2)
+    assert sh.comm_mode == "allgather", (sh.comm_mode, sh.halo_fraction)
+    assert sh.halo_fraction > 0.5
+    assert sh.ag_colidx is not None and sh.ag_t_rowptr is not None
+    # gather-space ids decode back to the right global ids
+    mr = sh.ag_max_rows
+    ag = sh.ag_colidx.to(torch.int64)
+    owner = ag // mr
+    local = ag % mr
+    bounds = torch.tensor(sh.bounds)
+    glob = bounds[owner] + local
+    # compare against ext-space remap: local cols < n_local are rank0's
+    rp = g.rowptr
+    want = g.colidx[rp[sh.lo]:rp[sh.hi]].to(torch.int64)
+    assert torch.equal(glob, want)
+
+
 def _worker(rank, port, model_name, q):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
-        os.environ["ROC_COMM_MODE"] = "halo"  # exercise the overlap path
         dist.init_process_group("gloo", rank=rank, world_size=WS)
         g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.08,
-                                                      seed=12)
+                                                      seed=13)
         bounds = edge_balanced_bounds(g.rowptr, WS)
-        sh = build_shard(g, rank, WS, bounds)
-        assert sh.has_overlap_split
         ws = {}
-        for overlap in ("0", "1"):
-            os.environ["ROC_OVERLAP"] = overlap
+        for mode in ("halo", "allgather"):
+            os.environ["ROC_COMM_MODE"] = mode
+            sh = build_shard(g, rank, WS, bounds)
+            assert sh.comm_mode == mode
             model = build_model(model_name, [feats.shape[1], 16, c],
                                 dropout=0.0, seed=1)
             opt = AdamOptimizer(model.parameters(), lr=0.01)
             tr = Trainer(model, sh, feats, labels, mask, opt)
             for _ in range(3):
                 tr.train_epoch()
-            key = "overlap" if overlap == "1" else "seq"
             p0 = next(iter(model.parameters()))
-            ws[key] = p0.detach().numpy().copy()
+            ws[mode] = p0.detach().numpy().copy()
         q.put((rank, ws, None))
     except Exception:  # pragma: no cover
         import traceback
         q.put((rank, None, traceback.format_exc()))
     finally:
-        os.environ.pop("ROC_OVERLAP", None)
         os.environ.pop("ROC_COMM_MODE", None)
         if dist.is_initialized():
             dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("model_name,port", [("gcn", 29561), ("sage", 29563),
-                                             ("gin", 29565)])
-def test_overlap_matches_sequential(model_name, port):
+@pytest.mark.parametrize("model_name,port", [("gcn", 29591), ("sage", 29593),
+                                             ("gin", 29595)])
+def test_allgather_matches_halo(model_name, port):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=_worker, args=(r, port, model_name, q))
@@ -63,7 +78,7 @@ def test_overlap_matches_sequential(model_name, port):
         p.join(timeout=180)
     for rank, ws, err in res:
         assert err is None, f"rank {rank}: {err}"
-        a = torch.from_numpy(ws["seq"])
-        b = torch.from_numpy(ws["overlap"])
+        a = torch.from_numpy(ws["halo"])
+        b = torch.from_numpy(ws["allgather"])
         assert torch.allclose(a, b, atol=1e-5), \
             (model_name, rank, (a - b).abs().max())
