@@ -45,6 +45,7 @@ def _worker(rank, port, lux_path, q):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
+        os.environ["ROC_COMM_MODE"] = "halo"  # compare the halo structures
         dist.init_process_group("gloo", rank=rank, world_size=WS)
         g = load_lux(lux_path)
         bounds = edge_balanced_bounds(g.rowptr, WS)
@@ -63,6 +64,7 @@ def _worker(rank, port, lux_path, q):
         import traceback
         q.put((rank, None, traceback.format_exc()))
     finally:
+        os.environ.pop("ROC_COMM_MODE", None)
         if dist.is_initialized():
             dist.destroy_process_group()
 
