@@ -159,7 +159,8 @@ def main():
                 "graph": f"{args.dataset}-synthetic-{g.num_nodes}n-{g.num_edges}e",
                 "global_batch": g.num_nodes,
                 "seq_len": None,
-                "parallelism": f"graph-partition x{world} (halo exchange)",
+                "parallelism": f"graph-partition x{world} "
+                               f"({shard.comm_mode if world > 1 else 'single-gpu'})",
             },
         }
         print(json.dumps(result), flush=True)
