@@ -291,3 +291,34 @@ def test_adam_gpu():
     assert torch.allclose(wd.cpu(), w, atol=1e-6)
     assert torch.allclose(md_.cpu(), m, atol=1e-6)
     assert torch.allclose(vd.cpu(), v, atol=1e-6)
+
+
+def test_spmm_accumulate_mode(graph_small):
+    """Two-pass aggregation (edge split + accumulate flag, the halo-overlap
+    building block) must equal the single-pass result."""
+    g, shard = graph_small
+    torch.manual_seed(14)
+    n = g.num_nodes
+    x = torch.randn(n, 64).to(torch.bfloat16).to(DEV)
+    rowptr = g.rowptr.to(DEV)
+    colidx = g.colidx.to(DEV)
+    deg = (g.rowptr[1:] - g.rowptr[:-1]).float().clamp(min=1)
+    dst = deg.rsqrt().to(DEV)
+    want = torch.empty_like(x)
+    _ext().spmm(want, x, rowptr, colidx, dst, None, None)
+    # split each row's edges at the midpoint into two CSRs
+    rp = g.rowptr
+    mid = (rp[:-1] + (rp[1:] - rp[:-1]) // 2)
+    rp1 = torch.zeros(n + 1, dtype=torch.int64)
+    rp1[1:] = torch.cumsum(mid - rp[:-1], 0)
+    rp2 = torch.zeros(n + 1, dtype=torch.int64)
+    rp2[1:] = torch.cumsum(rp[1:] - mid, 0)
+    ci1 = torch.cat([g.colidx[rp[v]:mid[v]] for v in range(n)])
+    ci2 = torch.cat([g.colidx[mid[v]:rp[v + 1]] for v in range(n)])
+    got = torch.empty_like(x)
+    _ext().spmm(got, x, rp1.to(DEV), ci1.to(DEV), None, None, None, False)
+    _ext().spmm(got, x, rp2.to(DEV), ci2.to(DEV), dst, None, None, True)
+    # pass-2 starts from bf16 partials: one extra rounding step
+    tol = want.float().abs().max().item() * 2 ** -6
+    assert torch.allclose(got.float(), want.float(), atol=tol, rtol=0.05), \
+        (got.float() - want.float()).abs().max()
