@@ -81,7 +81,8 @@ class _GatherAggregate(torch.autograd.Function):
         mr = shard.ag_max_rows
         ws = shard.world_size
         dfull = torch.empty(ws * mr, D, dtype=dy.dtype, device=dy.device)
-        _spmm_local(dfull, dy, shard.ag_t_rowptr, shard.ag_t_colidx, None)
+        _spmm_local(dfull, dy, shard.ag_t_rowptr, shard.ag_t_colidx, None,
+                    shard.ag_t_row_order if dy.is_cuda else None)
         dx_pad = torch.empty(mr, D, dtype=dy.dtype, device=dy.device)
         _reduce_scatter(dx_pad, dfull, group)
         return dx_pad[:shard.n_local].contiguous(), None, None, None

@@ -68,6 +68,7 @@ class GraphShard:
     ag_colidx: Optional[torch.Tensor] = None     # cols in gather space
     ag_t_rowptr: Optional[torch.Tensor] = None   # [ws*ag_max_rows+1]
     ag_t_colidx: Optional[torch.Tensor] = None
+    ag_t_row_order: Optional[torch.Tensor] = None
 
     def to(self, device) -> "GraphShard":
         d = {}
@@ -259,6 +260,9 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
         ag_t = build_transpose(world_size * mr, rowptr_t,
                                shard_kw["ag_colidx"])
         shard_kw["ag_t_rowptr"], shard_kw["ag_t_colidx"] = ag_t
+        ag_t_deg = np.diff(ag_t[0].numpy())
+        shard_kw["ag_t_row_order"] = torch.from_numpy(
+            np.argsort(-ag_t_deg, kind="stable").astype(np.int32))
     if world_size > 1 and shard_kw.get("comm_mode") == "halo":
         # split edges by source locality for comm/compute overlap
         is_loc_edge = colidx < n_local
