@@ -64,9 +64,11 @@ def main():
         dist.init_process_group("nccl" if on_gpu else "gloo",
                                 rank=rank, world_size=world)
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        # ROC_DEVICE_OVERRIDE: testing hook (e.g. 2 RCCL ranks on 1 GPU)
+        dev_idx = int(os.environ.get("ROC_DEVICE_OVERRIDE", local_rank))
         if on_gpu:
-            torch.cuda.set_device(local_rank)
-            device = f"cuda:{local_rank}"
+            torch.cuda.set_device(dev_idx)
+            device = f"cuda:{dev_idx}"
         else:
             device = "cpu"
     else:
