@@ -11,7 +11,6 @@ serially, `optimizer_kernel.cu:88-94`).
 from __future__ import annotations
 
 import time
-from typing import Optional
 
 import torch
 import torch.distributed as dist

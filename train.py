@@ -14,7 +14,6 @@ Reference flag parity: --file (.lux dataset prefix), --layers D0-D1-...-C,
 """
 import argparse
 import os
-import sys
 import time
 
 import torch
