@@ -51,10 +51,39 @@ __device__ __forceinline__ void acc_add(float* acc, const float4& r, float w) {
   acc[0] += w * r.x; acc[1] += w * r.y; acc[2] += w * r.z; acc[3] += w * r.w;
 }
 
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+
+// Two gather engines: plain global loads (64-bit addressing), and SRSRC
+// buffer loads with 32-bit offsets when the matrix fits 4 GB — fewer
+// address VGPRs = more waves in flight on a latency-bound kernel
+// (CDNA guide T8/T20; the descriptor is built from kernargs only, so no
+// waterfall loops).
+template <typename T>
+struct GlobalGather {
+  const T* x;
+  int64_t D;
+  int64_t col0;
+  __device__ __forceinline__ typename RawVec<T>::type load(int u) const {
+    return load_raw(x + (int64_t)u * D + col0);
+  }
+};
+
+template <typename T>
+struct BufferGather {
+  __amdgpu_buffer_rsrc_t rsrc;
+  unsigned dbytes;    // D * sizeof(T)
+  unsigned colbytes;  // col0 * sizeof(T)
+  __device__ __forceinline__ typename RawVec<T>::type load(int u) const {
+    const f32x4v v = __builtin_amdgcn_raw_buffer_load_b128(
+        rsrc, (unsigned)u * dbytes + colbytes, 0, 0);
+    return __builtin_bit_cast(typename RawVec<T>::type, v);
+  }
+};
+
 // hot path: this lane's 16-B unit is entirely inside D
-template <typename T, int EPU, bool UN8>
+template <typename T, int EPU, bool UN8, typename LD>
 __device__ __forceinline__ void row_accum_vec(
-    float* __restrict__ acc, const T* __restrict__ x, int64_t D, int64_t col0,
+    float* __restrict__ acc, const LD& ld,
     const int* __restrict__ colidx, const float* __restrict__ deg_src,
     int64_t e0, int64_t e1) {
   using Raw = typename RawVec<T>::type;
@@ -66,14 +95,14 @@ __device__ __forceinline__ void row_accum_vec(
       const int u2 = colidx[e + 2], u3 = colidx[e + 3];
       const int u4 = colidx[e + 4], u5 = colidx[e + 5];
       const int u6 = colidx[e + 6], u7 = colidx[e + 7];
-      const Raw r0 = load_raw(x + (int64_t)u0 * D + col0);
-      const Raw r1 = load_raw(x + (int64_t)u1 * D + col0);
-      const Raw r2 = load_raw(x + (int64_t)u2 * D + col0);
-      const Raw r3 = load_raw(x + (int64_t)u3 * D + col0);
-      const Raw r4 = load_raw(x + (int64_t)u4 * D + col0);
-      const Raw r5 = load_raw(x + (int64_t)u5 * D + col0);
-      const Raw r6 = load_raw(x + (int64_t)u6 * D + col0);
-      const Raw r7 = load_raw(x + (int64_t)u7 * D + col0);
+      const Raw r0 = ld.load(u0);
+      const Raw r1 = ld.load(u1);
+      const Raw r2 = ld.load(u2);
+      const Raw r3 = ld.load(u3);
+      const Raw r4 = ld.load(u4);
+      const Raw r5 = ld.load(u5);
+      const Raw r6 = ld.load(u6);
+      const Raw r7 = ld.load(u7);
       if (deg_src) {
         acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
         acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
@@ -90,10 +119,10 @@ __device__ __forceinline__ void row_accum_vec(
   for (; e + 3 < e1; e += 4) {
     const int u0 = colidx[e], u1 = colidx[e + 1];
     const int u2 = colidx[e + 2], u3 = colidx[e + 3];
-    const Raw r0 = load_raw(x + (int64_t)u0 * D + col0);
-    const Raw r1 = load_raw(x + (int64_t)u1 * D + col0);
-    const Raw r2 = load_raw(x + (int64_t)u2 * D + col0);
-    const Raw r3 = load_raw(x + (int64_t)u3 * D + col0);
+    const Raw r0 = ld.load(u0);
+    const Raw r1 = ld.load(u1);
+    const Raw r2 = ld.load(u2);
+    const Raw r3 = ld.load(u3);
     if (deg_src) {
       acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
       acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
@@ -104,7 +133,7 @@ __device__ __forceinline__ void row_accum_vec(
   }
   for (; e < e1; ++e) {
     const int u0 = colidx[e];
-    const Raw r0 = load_raw(x + (int64_t)u0 * D + col0);
+    const Raw r0 = ld.load(u0);
     acc_add(acc, r0, deg_src ? deg_src[u0] : 1.f);
   }
 }
@@ -124,13 +153,13 @@ __device__ void row_accum_tail(
   }
 }
 
-template <typename T, int TEAM, bool UN8>
+template <typename T, int TEAM, bool UN8, bool BUF>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
     T* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
     const float* __restrict__ deg_dst, const float* __restrict__ deg_src,
     const int* __restrict__ row_order, int num_rows, int64_t D,
-    bool accumulate) {
+    bool accumulate, unsigned x_bytes) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const int tpb = kBlock / TEAM;
   const int team = blockIdx.x * tpb + (int)threadIdx.x / TEAM;
@@ -155,7 +184,16 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
 #pragma unroll
         for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
       }
-      row_accum_vec<T, EPU, UN8>(acc, x, D, col0, colidx, deg_src, e0, e1);
+      if constexpr (BUF) {
+        BufferGather<T> ld{
+            __builtin_amdgcn_make_buffer_rsrc((void*)x, (short)0, x_bytes,
+                                              0x00020000),
+            (unsigned)(D * sizeof(T)), (unsigned)(col0 * sizeof(T))};
+        row_accum_vec<T, EPU, UN8>(acc, ld, colidx, deg_src, e0, e1);
+      } else {
+        GlobalGather<T> ld{x, D, col0};
+        row_accum_vec<T, EPU, UN8>(acc, ld, colidx, deg_src, e0, e1);
+      }
       if (deg_dst) {
         const float s = deg_dst[row];
 #pragma unroll
@@ -182,7 +220,7 @@ template <typename T>
 void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
                  const float* deg_dst, const float* deg_src,
                  const int* row_order, int num_rows, int64_t D,
-                 bool accumulate, hipStream_t stream) {
+                 bool accumulate, size_t x_elems, hipStream_t stream) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const int64_t units = (D + EPU - 1) / EPU;
   int team = 8;
@@ -193,28 +231,42 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   // read per call (cheap vs a ms-scale launch) so A/B harnesses can toggle
   const char* un8_env = getenv("ROC_SPMM_UNROLL8");
   const bool un8 = !(un8_env && un8_env[0] == '0');  // default ON
+  const char* buf_env = getenv("ROC_SPMM_BUFFER");
+  const size_t xb = x_elems * sizeof(T);
+  const bool buf = xb < (size_t)UINT_MAX && !(buf_env && buf_env[0] == '0');
+  const unsigned x_bytes = (unsigned)(buf ? xb : 0);
   // measured (scripts/bench_spmm.py, Reddit shape): degree-descending
   // scheduling wins for wide rows (D=256: -13%) but loses for narrow
   // ones (D=48: +10% — the indirection costs more than the skew tail)
   if (team < 16) row_order = nullptr;
-#define ROC_SPMM_LAUNCH(TEAM_, UN8_)                                        \
-  hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_>), grid, dim3(kBlock), 0,  \
-                     stream, out, x, rowptr, colidx, deg_dst, deg_src,      \
-                     row_order, num_rows, D, accumulate)
+#define ROC_SPMM_L2(TEAM_, UN8_)                                            \
+  do {                                                                      \
+    if (buf) {                                                              \
+      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, true>), grid,         \
+                         dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
+                         deg_dst, deg_src, row_order, num_rows, D,          \
+                         accumulate, x_bytes);                              \
+    } else {                                                                \
+      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, false>), grid,        \
+                         dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
+                         deg_dst, deg_src, row_order, num_rows, D,          \
+                         accumulate, x_bytes);                              \
+    }                                                                       \
+  } while (0)
   switch (team) {
     case 8:
-      if (un8) { ROC_SPMM_LAUNCH(8, true); } else { ROC_SPMM_LAUNCH(8, false); }
+      if (un8) { ROC_SPMM_L2(8, true); } else { ROC_SPMM_L2(8, false); }
       break;
     case 16:
-      if (un8) { ROC_SPMM_LAUNCH(16, true); } else { ROC_SPMM_LAUNCH(16, false); }
+      if (un8) { ROC_SPMM_L2(16, true); } else { ROC_SPMM_L2(16, false); }
       break;
     case 32:
-      if (un8) { ROC_SPMM_LAUNCH(32, true); } else { ROC_SPMM_LAUNCH(32, false); }
+      if (un8) { ROC_SPMM_L2(32, true); } else { ROC_SPMM_L2(32, false); }
       break;
     default:
-      if (un8) { ROC_SPMM_LAUNCH(64, true); } else { ROC_SPMM_LAUNCH(64, false); }
+      if (un8) { ROC_SPMM_L2(64, true); } else { ROC_SPMM_L2(64, false); }
   }
-#undef ROC_SPMM_LAUNCH
+#undef ROC_SPMM_L2
 }
 
 }  // namespace
@@ -245,11 +297,12 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
     launch_spmm<unsigned short>(
         (unsigned short*)out.data_ptr(), (const unsigned short*)x.data_ptr(),
         rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
-        num_rows, D, accumulate, stream);
+        num_rows, D, accumulate, (size_t)x.numel(), stream);
   } else if (x.scalar_type() == torch::kFloat32) {
     launch_spmm<float>(out.data_ptr<float>(), x.data_ptr<float>(),
                        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd,
-                       ds, ro, num_rows, D, accumulate, stream);
+                       ds, ro, num_rows, D, accumulate, (size_t)x.numel(),
+                       stream);
   } else {
     TORCH_CHECK(false, "spmm: unsupported dtype (bf16/f32 only)");
   }
