@@ -63,21 +63,21 @@ def main():
             print(f"  src-window {window:6d} rows ({mb:6.1f} MB): "
                   f"median {med:8.2f} ms ({gb/med:.2f} TB/s eff)", flush=True)
         variants = {}
-        for un8 in (0, 1):
+        os.environ["ROC_SPMM_UNROLL8"] = "1"
+        for use_buf in (0, 1):
             for use_order in (0, 1):
-                for use_src in (0, 1):
-                    os.environ["ROC_SPMM_UNROLL8"] = str(un8)
-                    ro = row_order if use_order else None
-                    ds = rsq if use_src else None
+                os.environ["ROC_SPMM_BUFFER"] = str(use_buf)
+                ro = row_order if use_order else None
 
-                    def fn(ro=ro, ds=ds):
-                        _C.spmm(out, x, rowptr, colidx, rsq, ds, ro)
+                def fn(ro=ro):
+                    _C.spmm(out, x, rowptr, colidx, rsq, None, ro)
 
-                    med, best = time_variant(fn, args.rounds)
-                    key = f"un8={un8} order={use_order} srcdeg={use_src}"
-                    variants[key] = (med, best)
-                    print(f"  {key}: median {med:8.2f} ms  best {best:8.2f}"
-                          f"  ({gb/med:.2f} TB/s eff)", flush=True)
+                med, best = time_variant(fn, args.rounds)
+                key = f"buf={use_buf} order={use_order}"
+                variants[key] = (med, best)
+                print(f"  {key}: median {med:8.2f} ms  best {best:8.2f}"
+                      f"  ({gb/med:.2f} TB/s eff)", flush=True)
+        os.environ.pop("ROC_SPMM_BUFFER", None)
     print("done")
 
 
