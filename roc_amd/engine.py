@@ -43,6 +43,7 @@ class Trainer:
         self.labels = labels[lo:hi].to(device=device).contiguous()
         self.mask = mask[lo:hi].to(device=device, dtype=torch.int32).contiguous()
         self.optimizer = optimizer
+        optimizer.setup_flat_grads()  # single-buffer grads (one all-reduce)
         self.num_classes = num_classes  # true classes if logits are padded
         self.epoch = 0
         F.set_dropout_seed(seed + shard.rank * 7919)
@@ -121,20 +122,11 @@ class Trainer:
         self.optimizer.t += 1
         return self._graph_metrics
 
-    # -- gradient all-reduce (flat bucket; weights are small) ---------------
+    # -- gradient all-reduce (flat buffer; weights are small) ---------------
     def _allreduce_grads(self):
         if self.shard.world_size == 1 or not dist.is_initialized():
             return
-        grads = [p.grad for p in self.optimizer.params if p.grad is not None]
-        if not grads:
-            return
-        flat = torch.cat([g.reshape(-1) for g in grads])
-        dist.all_reduce(flat, group=self.group)
-        off = 0
-        for g in grads:
-            n = g.numel()
-            g.copy_(flat[off:off + n].view_as(g))
-            off += n
+        dist.all_reduce(self.optimizer._flat_grad, group=self.group)
 
     offload = None  # set via enable_offload()
 

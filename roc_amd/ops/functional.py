@@ -160,7 +160,8 @@ class _Linear(torch.autograd.Function):
     def forward(ctx, x, w, act, row_scale):
         if _hip(x):
             # gemm_rr takes B pre-transposed: Bt = w^T [out,in]
-            wt = w.t().contiguous().to(x.dtype)
+            wt, wc = _cast_weight(w, x.dtype)
+            ctx.wc = wc
             y = torch.empty(x.shape[0], w.shape[1], dtype=x.dtype, device=x.device)
             _C.gemm_rr(y, x, wt, act == "relu", row_scale)
             if act == "sigmoid":
@@ -200,9 +201,8 @@ class _Linear(torch.autograd.Function):
             dx = None
             if ctx.needs_input_grad[0]:  # layer-1 inputs carry no grad
                 # dx = dy @ w^T: gemm_rr's Bt = (w^T)^T = w [in,out]
-                wc = w.contiguous().to(dy.dtype)
                 dx = torch.empty_like(x)
-                _C.gemm_rr(dx, dy, wc, False)
+                _C.gemm_rr(dx, dy, ctx.wc, False)
         else:
             if act == "relu":
                 dy = ref.relu_grad(dy, y)
@@ -312,6 +312,29 @@ def mul(a, b):
 # ---------------------------------------------------------------------------
 
 _DROPOUT_STATE = {"seed": 1, "offset": 0, "counter": None}
+_WEIGHT_VERSION = [0]
+
+
+def bump_weight_version() -> None:
+    """Called by the optimizer after a step: invalidates cached weight
+    casts (the per-call w -> bf16 (+transpose) casts are reused within a
+    step; disabled under hipGraph capture where the cast kernels must be
+    part of the replayed graph)."""
+    _WEIGHT_VERSION[0] += 1
+
+
+def _cast_weight(w, dtype):
+    if _DROPOUT_STATE["counter"] is not None:  # capture mode: no caching
+        wc = w.contiguous().to(dtype)
+        return wc.t().contiguous(), wc
+    cache = getattr(w, "_roc_cast", None)
+    ver = _WEIGHT_VERSION[0]
+    if cache is not None and cache[0] == ver and cache[2].dtype == dtype:
+        return cache[1], cache[2]
+    wc = w.contiguous().to(dtype)
+    wt = wc.t().contiguous()
+    w._roc_cast = (ver, wt, wc)
+    return wt, wc
 
 
 def set_dropout_seed(seed: int) -> None:

@@ -31,6 +31,22 @@ class AdamOptimizer:
         self.m = [torch.zeros_like(p.data) for p in self.params]
         self.v = [torch.zeros_like(p.data) for p in self.params]
         self._step_dev = None  # device schedule (hipGraph capture mode)
+        self._flat_grad = None
+
+    def setup_flat_grads(self) -> torch.Tensor:
+        """Point every param's .grad into ONE flat fp32 buffer: autograd
+        accumulates in place, the gradient all-reduce is a single
+        collective on the buffer (no cat/copy-back), and zero_grad is one
+        fill. Call after the model is on its final device."""
+        total = sum(p.numel() for p in self.params)
+        dev = self.params[0].device if self.params else "cpu"
+        self._flat_grad = torch.zeros(total, dtype=torch.float32, device=dev)
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            p.grad = self._flat_grad[off:off + n].view_as(p)
+            off += n
+        return self._flat_grad
 
     def set_device_step(self, step_tensor) -> None:
         """int64[1] device counter (bumped by the engine inside the
@@ -38,6 +54,9 @@ class AdamOptimizer:
         self._step_dev = step_tensor
 
     def zero_grad(self) -> None:
+        if self._flat_grad is not None:
+            self._flat_grad.zero_()
+            return
         for p in self.params:
             if p.grad is not None:
                 p.grad.detach_()
@@ -67,6 +86,7 @@ class AdamOptimizer:
             F.adam_step(p.data, p.grad.data, self.m[i], self.v[i], alpha_t,
                         self.beta1, self.beta2, self.eps, self.weight_decay,
                         **step_kw)
+        F.bump_weight_version()  # invalidate cached weight casts
 
     def state_dict(self) -> dict:
         return {"t": self.t, "m": self.m, "v": self.v,
