@@ -80,8 +80,11 @@ struct BufferGather {
   }
 };
 
-// hot path: this lane's 16-B unit is entirely inside D
-template <typename T, int EPU, bool UN8, typename LD>
+// hot path: this lane's 16-B unit is entirely inside D.
+// SRC is compile-time: the fused GCN path has no per-edge source scale,
+// and dropping the dead deg_src lookups frees the 8 index registers
+// (they otherwise stay live across all 8 in-flight gathers).
+template <typename T, int EPU, bool UN8, bool SRC, typename LD>
 __device__ __forceinline__ void row_accum_vec(
     float* __restrict__ acc, const LD& ld,
     const int* __restrict__ colidx, const float* __restrict__ deg_src,
@@ -103,7 +106,7 @@ __device__ __forceinline__ void row_accum_vec(
       const Raw r5 = ld.load(u5);
       const Raw r6 = ld.load(u6);
       const Raw r7 = ld.load(u7);
-      if (deg_src) {
+      if (SRC) {
         acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
         acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
         acc_add(acc, r4, deg_src[u4]); acc_add(acc, r5, deg_src[u5]);
@@ -123,7 +126,7 @@ __device__ __forceinline__ void row_accum_vec(
     const Raw r1 = ld.load(u1);
     const Raw r2 = ld.load(u2);
     const Raw r3 = ld.load(u3);
-    if (deg_src) {
+    if (SRC) {
       acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
       acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
     } else {
@@ -134,7 +137,7 @@ __device__ __forceinline__ void row_accum_vec(
   for (; e < e1; ++e) {
     const int u0 = colidx[e];
     const Raw r0 = ld.load(u0);
-    acc_add(acc, r0, deg_src ? deg_src[u0] : 1.f);
+    acc_add(acc, r0, SRC ? deg_src[u0] : 1.f);
   }
 }
 
@@ -153,7 +156,7 @@ __device__ void row_accum_tail(
   }
 }
 
-template <typename T, int TEAM, bool UN8, bool BUF>
+template <typename T, int TEAM, bool UN8, bool BUF, bool SRC>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
     T* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
@@ -189,10 +192,10 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
             __builtin_amdgcn_make_buffer_rsrc((void*)x, (short)0, x_bytes,
                                               0x00020000),
             (unsigned)(D * sizeof(T)), (unsigned)(col0 * sizeof(T))};
-        row_accum_vec<T, EPU, UN8>(acc, ld, colidx, deg_src, e0, e1);
+        row_accum_vec<T, EPU, UN8, SRC>(acc, ld, colidx, deg_src, e0, e1);
       } else {
         GlobalGather<T> ld{x, D, col0};
-        row_accum_vec<T, EPU, UN8>(acc, ld, colidx, deg_src, e0, e1);
+        row_accum_vec<T, EPU, UN8, SRC>(acc, ld, colidx, deg_src, e0, e1);
       }
       if (deg_dst) {
         const float s = deg_dst[row];
@@ -239,19 +242,24 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   // scheduling wins for wide rows (D=256: -13%) but loses for narrow
   // ones (D=48: +10% — the indirection costs more than the skew tail)
   if (team < 16) row_order = nullptr;
-#define ROC_SPMM_L2(TEAM_, UN8_)                                            \
+#define ROC_SPMM_L3(TEAM_, UN8_, BUF_)                                      \
   do {                                                                      \
-    if (buf) {                                                              \
-      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, true>), grid,         \
+    if (deg_src) {                                                          \
+      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, BUF_, true>), grid,   \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
                          deg_dst, deg_src, row_order, num_rows, D,          \
                          accumulate, x_bytes);                              \
     } else {                                                                \
-      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, false>), grid,        \
+      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, BUF_, false>), grid,  \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
                          deg_dst, deg_src, row_order, num_rows, D,          \
                          accumulate, x_bytes);                              \
     }                                                                       \
+  } while (0)
+#define ROC_SPMM_L2(TEAM_, UN8_)                                            \
+  do {                                                                      \
+    if (buf) { ROC_SPMM_L3(TEAM_, UN8_, true); }                            \
+    else     { ROC_SPMM_L3(TEAM_, UN8_, false); }                           \
   } while (0)
   switch (team) {
     case 8:
@@ -267,6 +275,7 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
       if (un8) { ROC_SPMM_L2(64, true); } else { ROC_SPMM_L2(64, false); }
   }
 #undef ROC_SPMM_L2
+#undef ROC_SPMM_L3
 }
 
 }  // namespace
