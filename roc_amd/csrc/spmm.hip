@@ -156,6 +156,9 @@ __device__ void row_accum_tail(
   }
 }
 
+// (a forced 6-waves/SIMD __launch_bounds__ hint was measured NEUTRAL:
+//  the allocator spills 28 B/lane on half the hot variants and any
+//  occupancy gain washes out — keep the default allocation)
 template <typename T, int TEAM, bool UN8, bool BUF, bool SRC>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
     T* __restrict__ out, const T* __restrict__ x,
