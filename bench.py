@@ -162,6 +162,9 @@ def main():
                 "seq_len": None,
                 "parallelism": f"graph-partition x{world} "
                                f"({shard.comm_mode if world > 1 else 'single-gpu'})",
+                "note": "dims padded for alignment (602->608 feats zero-"
+                        "cols, 41->64 classes zero-frozen): padded work is"
+                        " computed, never skipped; softmax uses true 41",
             },
         }
         print(json.dumps(result), flush=True)
