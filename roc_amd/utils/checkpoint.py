@@ -57,4 +57,5 @@ def load_checkpoint(path: str, trainer) -> dict:
     F._DROPOUT_STATE.update(state["dropout_state"])
     F._DROPOUT_STATE["counter"] = counter  # device tensor is not persisted
     torch.set_rng_state(state["torch_rng"])
+    F.bump_weight_version()  # invalidate cached weight casts (in-place load)
     return state.get("extra", {})
