@@ -190,6 +190,31 @@ class Trainer:
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 
+    # -- reference train-verb API (gnn.h:162-203 parity) --------------------
+    # train_epoch() is the packaged loop; these verbs expose the same
+    # steps individually, in the reference's vocabulary.
+    def train_mode(self):
+        self.model.train()
+
+    def infer_mode(self):
+        self.model.eval()
+
+    def zero_gradients(self):
+        self.optimizer.zero_grad()
+
+    def forward(self):
+        self._last_loss, self._last_metrics = self._forward_loss()
+        return self._last_metrics
+
+    def backward(self):
+        if self.offload is not None:
+            self.offload.prefetch()
+        self._last_loss.backward()
+
+    def update(self):
+        self._allreduce_grads()
+        self.optimizer.step()
+
     # -- cost-model repartitioning (the MLSys'20 Roc idea; the reference
     #    code only has the static edge-balanced split) --------------------
     def measure_and_rebalance(self, feats, labels, mask, probe_epochs=3):

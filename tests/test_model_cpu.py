@@ -70,3 +70,26 @@ def test_eval_cadence_loop():
         if (epoch + 1) % 5 == 0:
             history.append(tr.evaluate())
     assert len(history) == 2
+
+
+def test_reference_verb_api():
+    """The reference's train verbs (gnn.h:162-203) drive the same loop as
+    train_epoch."""
+    tr_a = make_trainer("gcn")
+    from roc_amd.ops import functional as F
+    F.set_dropout_seed(77)
+    for _ in range(3):
+        tr_a.train_epoch()
+    wa = tr_a.model.weights[0].detach().clone()
+
+    tr_b = make_trainer("gcn")
+    F.set_dropout_seed(77)
+    for _ in range(3):
+        tr_b.train_mode()
+        tr_b.zero_gradients()
+        tr_b.forward()
+        tr_b.backward()
+        tr_b.update()
+    tr_b.infer_mode()
+    wb = tr_b.model.weights[0].detach()
+    assert torch.allclose(wa, wb, atol=1e-7)
