@@ -71,8 +71,14 @@ class _SpMM(torch.autograd.Function):
         dy = dy.contiguous()
         # d/dx of  diag(a) A diag(b) x  =  diag(b) A^T diag(a) dy
         if _hip(dy):
+            if deg_dst is not None:
+                # pre-scale the gradient rows ONCE (one cheap pass) instead
+                # of a per-edge deg gather inside the transpose SpMM
+                tmp = torch.empty_like(dy)
+                _C.rowscale(tmp, dy, deg_dst)
+                dy = tmp
             dx = torch.empty(ctx.num_ext, dy.shape[1], dtype=dy.dtype, device=dy.device)
-            _C.spmm(dx, dy, t_rowptr, t_colidx, deg_src, deg_dst, t_row_order)
+            _C.spmm(dx, dy, t_rowptr, t_colidx, deg_src, None, t_row_order)
         else:
             yin = dy
             if deg_dst is not None:
