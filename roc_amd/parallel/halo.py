@@ -29,6 +29,9 @@ class _HaloExchange(torch.autograd.Function):
         ctx.group = group
         if shard.world_size == 1:
             return x
+        assert shard.comm_mode == "halo", (
+            "halo_exchange needs a halo-mode shard (this one is "
+            f"'{shard.comm_mode}'; use parallel.aggregate.aggregate())")
         # NOTE: the all_to_all is collective — every rank participates even
         # with an empty halo (another rank may still need our rows).
         n_local = shard.n_local

@@ -201,10 +201,20 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
     colidx_t = torch.from_numpy(colidx)
     t_rowptr, t_colidx = build_transpose(n_ext, rowptr_t, colidx_t)
 
-    # who needs MY rows
-    if world_size == 1:
+    # exchange strategy (parallel/aggregate.py dispatches on it)
+    comm_mode = "none"
+    halo_fraction = 0.0
+    if world_size > 1:
+        import os as _os
+        halo_fraction = n_halo / max(bounds[-1] - n_local, 1)
+        comm_mode = _os.environ.get("ROC_COMM_MODE", "auto")
+        if comm_mode == "auto":
+            comm_mode = "allgather" if halo_fraction > 0.5 else "halo"
+
+    # who needs MY rows (halo mode only; allgather sends whole blocks)
+    if world_size == 1 or comm_mode == "allgather":
         send_idx = torch.empty(0, dtype=torch.int64)
-        send_splits = [0]
+        send_splits = [0] * world_size
     elif full_colidx is not None:
         # scan every rank's window (no process group needed; deterministic)
         ci = full_colidx.numpy()
@@ -239,15 +249,9 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
 
     shard_kw = {}
     if world_size > 1:
-        num_nodes_total = bounds[-1]
-        halo_fraction = n_halo / max(num_nodes_total - n_local, 1)
-        import os as _os
-        mode = _os.environ.get("ROC_COMM_MODE", "auto")
-        if mode == "auto":
-            mode = "allgather" if halo_fraction > 0.5 else "halo"
-        shard_kw["comm_mode"] = mode
+        shard_kw["comm_mode"] = comm_mode
         shard_kw["halo_fraction"] = float(halo_fraction)
-    if world_size > 1 and shard_kw.get("comm_mode") == "allgather":
+    if world_size > 1 and comm_mode == "allgather":
         # gather-space CSR: node (owner r, local i) -> r*max_rows + i
         sizes = [bounds[r + 1] - bounds[r] for r in range(world_size)]
         mr = max(sizes)
@@ -263,7 +267,7 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
         ag_t_deg = np.diff(ag_t[0].numpy())
         shard_kw["ag_t_row_order"] = torch.from_numpy(
             np.argsort(-ag_t_deg, kind="stable").astype(np.int32))
-    if world_size > 1 and shard_kw.get("comm_mode") == "halo":
+    if world_size > 1 and comm_mode == "halo":
         # split edges by source locality for comm/compute overlap
         is_loc_edge = colidx < n_local
         row_of_edge = np.repeat(np.arange(n_local, dtype=np.int64),
@@ -274,9 +278,9 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
             if key_pfx == "halo":
                 cols_sel = cols_sel - n_local  # halo-local id space
             cnt = np.bincount(rows_sel, minlength=n_local).astype(np.int64)
-            rp = np.zeros(n_local + 1, dtype=np.int64)
-            np.cumsum(cnt, out=rp[1:])
-            shard_kw[f"{key_pfx}_rowptr"] = torch.from_numpy(rp)
+            part_rp = np.zeros(n_local + 1, dtype=np.int64)
+            np.cumsum(cnt, out=part_rp[1:])
+            shard_kw[f"{key_pfx}_rowptr"] = torch.from_numpy(part_rp)
             shard_kw[f"{key_pfx}_colidx"] = torch.from_numpy(
                 np.ascontiguousarray(cols_sel))
         t_loc = build_transpose(n_local, shard_kw["loc_rowptr"],

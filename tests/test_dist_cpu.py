@@ -18,9 +18,12 @@ from roc_amd import build_model, AdamOptimizer, Trainer
 WS = 2
 
 
-def _init(rank, port):
+def _init(rank, port, comm_mode="halo"):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
+    # pin the exchange strategy: these tests exercise the halo plan
+    # directly (auto picks allgather on uniform synthetic graphs)
+    os.environ["ROC_COMM_MODE"] = comm_mode
     dist.init_process_group("gloo", rank=rank, world_size=WS)
 
 
