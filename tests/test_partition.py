@@ -41,7 +41,8 @@ def test_sharded_spmm_matches_global(world_size):
         assert torch.allclose(outn_local, outn_global[sh.lo:sh.hi], atol=1e-5)
 
 
-def test_send_recv_plans_agree():
+def test_send_recv_plans_agree(monkeypatch):
+    monkeypatch.setenv("ROC_COMM_MODE", "halo")  # send plans are halo-mode
     n, ws = 150, 3
     g = synthetic_graph(n, 2000, seed=13)
     bounds = edge_balanced_bounds(g.rowptr, ws)
