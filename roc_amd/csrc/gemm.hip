@@ -327,6 +327,8 @@ void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu,
   const int N = (int)Bt.size(0);
   TORCH_CHECK(Bt.size(1) == K, "gemm_rr: Bt must be [N,K]");
   TORCH_CHECK(C.size(0) == M && C.size(1) == N, "gemm_rr: C shape");
+  if (M == 0 || N == 0) return;  // empty partition
+  if (K == 0) { C.zero_(); return; }
   auto s = roc_stream();
   const float* rs = nullptr;
   if (row_scale.has_value()) {
@@ -363,6 +365,7 @@ void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B) {
   const int N = (int)B.size(1);
   TORCH_CHECK(B.size(0) == R, "gemm_atb: row mismatch");
   TORCH_CHECK(C.size(0) == Ka && C.size(1) == N, "gemm_atb: C shape");
+  if (R == 0 || Ka == 0 || N == 0) return;  // empty partition: C += 0
   auto s = roc_stream();
   auto* c = C.data_ptr<float>();
   if (A.scalar_type() == torch::kBFloat16) {

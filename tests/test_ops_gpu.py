@@ -322,3 +322,32 @@ def test_spmm_accumulate_mode(graph_small):
     tol = want.float().abs().max().item() * 2 ** -6
     assert torch.allclose(got.float(), want.float(), atol=tol, rtol=0.05), \
         (got.float() - want.float()).abs().max()
+
+
+def test_empty_and_odd_shapes():
+    """Degenerate partitions (0 rows) and odd dims must not fault."""
+    A0 = torch.empty(0, 32, dtype=torch.bfloat16, device=DEV)
+    B = torch.randn(16, 32, dtype=torch.bfloat16, device=DEV)
+    C0 = torch.empty(0, 16, dtype=torch.bfloat16, device=DEV)
+    _ext().gemm_rr(C0, A0, B, False)
+    dw = torch.zeros(32, 8, dtype=torch.float32, device=DEV)
+    _ext().gemm_atb(dw, A0, torch.empty(0, 8, dtype=torch.bfloat16,
+                                        device=DEV))
+    assert dw.abs().sum().item() == 0.0
+    torch.cuda.synchronize()
+
+
+def test_spmm_shape_sweep(graph_small):
+    """Kernel dispatch sweep: every (dtype, team-size, tail) combination
+    vs the torch reference on a real graph."""
+    g, shard = graph_small
+    shd = shard.to(DEV)
+    for D in (8, 16, 24, 48, 63, 64, 65, 120, 128, 200, 256, 320, 513):
+        for dt in (torch.float32, torch.bfloat16):
+            x = torch.randn(g.num_nodes, D).to(dt)
+            want = ref.spmm(x.float(), g.rowptr, g.colidx, g.num_nodes)
+            got = F.scatter_gather(x.to(DEV), shd).float().cpu()
+            tol = want.abs().max().item() * (2 ** -7 if dt == torch.bfloat16
+                                             else 1e-5) + 1e-3
+            assert torch.allclose(got, want, atol=tol, rtol=0.05), \
+                (D, dt, (got - want).abs().max())
