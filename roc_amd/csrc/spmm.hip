@@ -84,15 +84,55 @@ struct BufferGather {
 // SRC is compile-time: the fused GCN path has no per-edge source scale,
 // and dropping the dead deg_src lookups frees the 8 index registers
 // (they otherwise stay live across all 8 in-flight gathers).
-template <typename T, int EPU, bool UN8, bool SRC, typename LD>
+template <typename T, int EPU, int UN, bool SRC, typename LD>
 __device__ __forceinline__ void row_accum_vec(
     float* __restrict__ acc, const LD& ld,
     const int* __restrict__ colidx, const float* __restrict__ deg_src,
     int64_t e0, int64_t e1) {
   using Raw = typename RawVec<T>::type;
   int64_t e = e0;
-  if (UN8) {
-    // 8 independent gathers in flight (raw payloads, unpack at use)
+  if (UN >= 16) {
+    // 16 in flight: two 8-groups issued back-to-back
+    for (; e + 15 < e1; e += 16) {
+      const int u0 = colidx[e], u1 = colidx[e + 1];
+      const int u2 = colidx[e + 2], u3 = colidx[e + 3];
+      const int u4 = colidx[e + 4], u5 = colidx[e + 5];
+      const int u6 = colidx[e + 6], u7 = colidx[e + 7];
+      const int v0 = colidx[e + 8], v1 = colidx[e + 9];
+      const int v2 = colidx[e + 10], v3 = colidx[e + 11];
+      const int v4 = colidx[e + 12], v5 = colidx[e + 13];
+      const int v6 = colidx[e + 14], v7 = colidx[e + 15];
+      const Raw r0 = ld.load(u0), r1 = ld.load(u1);
+      const Raw r2 = ld.load(u2), r3 = ld.load(u3);
+      const Raw r4 = ld.load(u4), r5 = ld.load(u5);
+      const Raw r6 = ld.load(u6), r7 = ld.load(u7);
+      const Raw s0 = ld.load(v0), s1 = ld.load(v1);
+      const Raw s2 = ld.load(v2), s3 = ld.load(v3);
+      const Raw s4 = ld.load(v4), s5 = ld.load(v5);
+      const Raw s6 = ld.load(v6), s7 = ld.load(v7);
+      if (SRC) {
+        acc_add(acc, r0, deg_src[u0]); acc_add(acc, r1, deg_src[u1]);
+        acc_add(acc, r2, deg_src[u2]); acc_add(acc, r3, deg_src[u3]);
+        acc_add(acc, r4, deg_src[u4]); acc_add(acc, r5, deg_src[u5]);
+        acc_add(acc, r6, deg_src[u6]); acc_add(acc, r7, deg_src[u7]);
+        acc_add(acc, s0, deg_src[v0]); acc_add(acc, s1, deg_src[v1]);
+        acc_add(acc, s2, deg_src[v2]); acc_add(acc, s3, deg_src[v3]);
+        acc_add(acc, s4, deg_src[v4]); acc_add(acc, s5, deg_src[v5]);
+        acc_add(acc, s6, deg_src[v6]); acc_add(acc, s7, deg_src[v7]);
+      } else {
+        acc_add(acc, r0, 1.f); acc_add(acc, r1, 1.f);
+        acc_add(acc, r2, 1.f); acc_add(acc, r3, 1.f);
+        acc_add(acc, r4, 1.f); acc_add(acc, r5, 1.f);
+        acc_add(acc, r6, 1.f); acc_add(acc, r7, 1.f);
+        acc_add(acc, s0, 1.f); acc_add(acc, s1, 1.f);
+        acc_add(acc, s2, 1.f); acc_add(acc, s3, 1.f);
+        acc_add(acc, s4, 1.f); acc_add(acc, s5, 1.f);
+        acc_add(acc, s6, 1.f); acc_add(acc, s7, 1.f);
+      }
+    }
+  }
+  if (UN >= 8) {
+    // 8 in flight (raw payloads, unpack at use)
     for (; e + 7 < e1; e += 8) {
       const int u0 = colidx[e], u1 = colidx[e + 1];
       const int u2 = colidx[e + 2], u3 = colidx[e + 3];
@@ -159,7 +199,7 @@ __device__ void row_accum_tail(
 // (a forced 6-waves/SIMD __launch_bounds__ hint was measured NEUTRAL:
 //  the allocator spills 28 B/lane on half the hot variants and any
 //  occupancy gain washes out — keep the default allocation)
-template <typename T, int TEAM, bool UN8, bool BUF, bool SRC>
+template <typename T, int TEAM, int UN, bool BUF, bool SRC>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
     T* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
@@ -195,10 +235,10 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
             __builtin_amdgcn_make_buffer_rsrc((void*)x, (short)0, x_bytes,
                                               0x00020000),
             (unsigned)(D * sizeof(T)), (unsigned)(col0 * sizeof(T))};
-        row_accum_vec<T, EPU, UN8, SRC>(acc, ld, colidx, deg_src, e0, e1);
+        row_accum_vec<T, EPU, UN, SRC>(acc, ld, colidx, deg_src, e0, e1);
       } else {
         GlobalGather<T> ld{x, D, col0};
-        row_accum_vec<T, EPU, UN8, SRC>(acc, ld, colidx, deg_src, e0, e1);
+        row_accum_vec<T, EPU, UN, SRC>(acc, ld, colidx, deg_src, e0, e1);
       }
       if (deg_dst) {
         const float s = deg_dst[row];
@@ -235,8 +275,8 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   const int tpb = kBlock / team;
   dim3 grid(roc_grid_1d(num_rows, tpb, 8192), col_tiles);
   // read per call (cheap vs a ms-scale launch) so A/B harnesses can toggle
-  const char* un8_env = getenv("ROC_SPMM_UNROLL8");
-  const bool un8 = !(un8_env && un8_env[0] == '0');  // default ON
+  const char* un_env = getenv("ROC_SPMM_UNROLL");
+  const int un = un_env ? atoi(un_env) : 8;  // 4 | 8 | 16, default 8
   const char* buf_env = getenv("ROC_SPMM_BUFFER");
   const size_t xb = x_elems * sizeof(T);
   const bool buf = xb < (size_t)UINT_MAX && !(buf_env && buf_env[0] == '0');
@@ -245,37 +285,37 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   // scheduling wins for wide rows (D=256: -13%) but loses for narrow
   // ones (D=48: +10% — the indirection costs more than the skew tail)
   if (team < 16) row_order = nullptr;
-#define ROC_SPMM_L3(TEAM_, UN8_, BUF_)                                      \
+#define ROC_SPMM_L3(TEAM_, UN_, BUF_)                                       \
   do {                                                                      \
     if (deg_src) {                                                          \
-      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, BUF_, true>), grid,   \
+      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN_, BUF_, true>), grid,    \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
                          deg_dst, deg_src, row_order, num_rows, D,          \
                          accumulate, x_bytes);                              \
     } else {                                                                \
-      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN8_, BUF_, false>), grid,  \
+      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN_, BUF_, false>), grid,   \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
                          deg_dst, deg_src, row_order, num_rows, D,          \
                          accumulate, x_bytes);                              \
     }                                                                       \
   } while (0)
-#define ROC_SPMM_L2(TEAM_, UN8_)                                            \
+#define ROC_SPMM_L2(TEAM_)                                                  \
   do {                                                                      \
-    if (buf) { ROC_SPMM_L3(TEAM_, UN8_, true); }                            \
-    else     { ROC_SPMM_L3(TEAM_, UN8_, false); }                           \
+    if (buf) {                                                              \
+      if (un >= 16) { ROC_SPMM_L3(TEAM_, 16, true); }                       \
+      else if (un >= 8) { ROC_SPMM_L3(TEAM_, 8, true); }                    \
+      else { ROC_SPMM_L3(TEAM_, 4, true); }                                 \
+    } else {                                                                \
+      if (un >= 16) { ROC_SPMM_L3(TEAM_, 16, false); }                      \
+      else if (un >= 8) { ROC_SPMM_L3(TEAM_, 8, false); }                   \
+      else { ROC_SPMM_L3(TEAM_, 4, false); }                                \
+    }                                                                       \
   } while (0)
   switch (team) {
-    case 8:
-      if (un8) { ROC_SPMM_L2(8, true); } else { ROC_SPMM_L2(8, false); }
-      break;
-    case 16:
-      if (un8) { ROC_SPMM_L2(16, true); } else { ROC_SPMM_L2(16, false); }
-      break;
-    case 32:
-      if (un8) { ROC_SPMM_L2(32, true); } else { ROC_SPMM_L2(32, false); }
-      break;
-    default:
-      if (un8) { ROC_SPMM_L2(64, true); } else { ROC_SPMM_L2(64, false); }
+    case 8:  ROC_SPMM_L2(8);  break;
+    case 16: ROC_SPMM_L2(16); break;
+    case 32: ROC_SPMM_L2(32); break;
+    default: ROC_SPMM_L2(64);
   }
 #undef ROC_SPMM_L2
 #undef ROC_SPMM_L3

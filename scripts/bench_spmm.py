@@ -63,21 +63,20 @@ def main():
             print(f"  src-window {window:6d} rows ({mb:6.1f} MB): "
                   f"median {med:8.2f} ms ({gb/med:.2f} TB/s eff)", flush=True)
         variants = {}
-        os.environ["ROC_SPMM_UNROLL8"] = "1"
-        for use_buf in (0, 1):
+        for un in (8, 16):
             for use_order in (0, 1):
-                os.environ["ROC_SPMM_BUFFER"] = str(use_buf)
+                os.environ["ROC_SPMM_UNROLL"] = str(un)
                 ro = row_order if use_order else None
 
                 def fn(ro=ro):
                     _C.spmm(out, x, rowptr, colidx, rsq, None, ro)
 
                 med, best = time_variant(fn, args.rounds)
-                key = f"buf={use_buf} order={use_order}"
+                key = f"un={un} order={use_order}"
                 variants[key] = (med, best)
                 print(f"  {key}: median {med:8.2f} ms  best {best:8.2f}"
                       f"  ({gb/med:.2f} TB/s eff)", flush=True)
-        os.environ.pop("ROC_SPMM_BUFFER", None)
+        os.environ.pop("ROC_SPMM_UNROLL", None)
     print("done")
 
 
