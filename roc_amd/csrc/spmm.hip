@@ -271,6 +271,11 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
   const int64_t units = (D + EPU - 1) / EPU;
   int team = 8;
   while (team < units && team < 64) team *= 2;
+  const char* team_env = getenv("ROC_SPMM_TEAM");  // geometry experiments
+  if (team_env) {
+    const int t = atoi(team_env);
+    if (t == 8 || t == 16 || t == 32 || t == 64) team = t;
+  }
   const int col_tiles = (int)((units + team - 1) / team);
   const int tpb = kBlock / team;
   dim3 grid(roc_grid_1d(num_rows, tpb, 8192), col_tiles);
