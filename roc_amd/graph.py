@@ -279,10 +279,37 @@ DATASET_SHAPES = {
 }
 
 
-def synthetic_dataset(name: str, seed: int = 1, scale: float = 1.0):
+def _learnable_labels(g: CSRGraph, feats: torch.Tensor, c: int,
+                      seed: int) -> torch.Tensor:
+    """Labels that are a (noisy) function of graph + features — a random
+    one-hop GCN 'teacher' — so full-scale convergence is demonstrable
+    (uniform random labels carry no learnable signal)."""
+    rng = np.random.default_rng(seed + 7)
+    d = feats.shape[1]
+    proj = torch.from_numpy(
+        (rng.standard_normal((d, c)) / np.sqrt(d)).astype(np.float32))
+    z = feats @ proj
+    agg = torch.zeros_like(z)
+    rp = g.rowptr.numpy()
+    dst_all = torch.repeat_interleave(
+        torch.arange(g.num_nodes, dtype=torch.int64),
+        torch.from_numpy(np.diff(rp)))
+    ci = g.colidx.to(torch.int64)
+    step = 20_000_000  # chunk the 10^8-edge gather
+    for s0 in range(0, g.num_edges, step):
+        s1 = min(g.num_edges, s0 + step)
+        agg.index_add_(0, dst_all[s0:s1], z[ci[s0:s1]])
+    score = z + agg * g.indegree().rsqrt().unsqueeze(1)
+    return score.argmax(dim=1)
+
+
+def synthetic_dataset(name: str, seed: int = 1, scale: float = 1.0,
+                      learnable_labels: bool = False):
     """Graph + features + labels + masks of the named shape.
 
     ``scale`` < 1 shrinks nodes/edges proportionally (for tests).
+    ``learnable_labels`` draws labels from a random one-hop teacher
+    instead of uniformly (same shapes/cost; lets accuracy actually rise).
     Returns (graph, features fp32 [N, in_dim], labels int64 [N], mask int32 [N]).
     """
     n, e, d, c = DATASET_SHAPES[name]
@@ -291,7 +318,10 @@ def synthetic_dataset(name: str, seed: int = 1, scale: float = 1.0):
     g = synthetic_graph(n, e, seed=seed)
     rng = np.random.default_rng(seed + 1)
     feats = torch.from_numpy(rng.standard_normal((n, d), dtype=np.float32))
-    labels = torch.from_numpy(rng.integers(0, c, size=n).astype(np.int64))
+    if learnable_labels:
+        labels = _learnable_labels(g, feats, c, seed)
+    else:
+        labels = torch.from_numpy(rng.integers(0, c, size=n).astype(np.int64))
     # 70/15/15 split like common full-graph benchmarks
     u = rng.random(n)
     mask = np.full(n, MASK_TRAIN, dtype=np.int32)

@@ -64,6 +64,9 @@ def parse_args():
                     help="write a chrome trace JSON here at the end")
     ap.add_argument("--offload", action="store_true",
                     help="host-DRAM activation offload (capacity tier)")
+    ap.add_argument("--learnable-labels", action="store_true",
+                    help="synthetic labels from a random one-hop teacher "
+                         "(accuracy can actually rise)")
     ap.add_argument("--rebalance-every", type=int, default=0,
                     help="cost-model repartition every N epochs (measured "
                          "per-rank throughput; multi-rank only)")
@@ -117,7 +120,8 @@ def load_dataset(args, rank=0, world=1):
         return (g, feats, labels, mask, dims[-1]), None
     name = args.dataset.replace("-synthetic", "")
     assert name in DATASET_SHAPES, f"unknown dataset {name}"
-    return synthetic_dataset(name, seed=args.seed, scale=args.scale), None
+    return synthetic_dataset(name, seed=args.seed, scale=args.scale,
+                             learnable_labels=args.learnable_labels), None
 
 
 def main():
