@@ -229,9 +229,16 @@ def synthetic_graph(
     seed: int = 1,
     skew: float = 1.0,
     add_self_edges: bool = True,
+    locality: float = 0.0,
+    num_communities: int = 64,
 ) -> CSRGraph:
     """Random directed graph with lognormal in-degree skew, self-edges added.
 
+    ``locality`` in [0,1): that fraction of edges stays inside the node's
+    own contiguous community block (planted-partition structure — real
+    graphs are community-heavy, and contiguous communities make the
+    vertex partition's halo SPARSE, exercising the halo/a2a strategy
+    instead of allgather).
     Column ids are sorted within each row (better gather locality, and the
     reference's CSR is sorted the same way after construction).
     """
@@ -249,6 +256,16 @@ def synthetic_graph(
     np.cumsum(deg + (1 if add_self_edges else 0), out=rowptr[1:])
     total = int(rowptr[-1])
     colidx = rng.integers(0, n, size=total, dtype=np.int64)
+    if locality > 0.0:
+        k = max(int(num_communities), 1)
+        block = (n + k - 1) // k
+        row_of_edge_l = np.repeat(np.arange(n, dtype=np.int64),
+                                  np.diff(rowptr))
+        local_sel = rng.random(total) < locality
+        base = (row_of_edge_l // block) * block
+        width = np.minimum(base + block, n) - base
+        colidx[local_sel] = (base[local_sel] +
+                             (colidx[local_sel] % width[local_sel]))
     if add_self_edges:
         # overwrite one slot per row with the self edge; then sort rows
         colidx[rowptr[:-1]] = np.arange(n, dtype=np.int64)

@@ -62,3 +62,21 @@ def test_indegree_clamped():
     g = synthetic_graph(30, 100, seed=5, add_self_edges=False)
     deg = g.indegree()
     assert (deg >= 1.0).all()
+
+
+def test_community_locality():
+    g = synthetic_graph(1000, 30000, seed=2, locality=0.9,
+                        num_communities=10)
+    rp, ci = g.rowptr.numpy(), g.colidx.numpy()
+    row = np.repeat(np.arange(1000), np.diff(rp))
+    same_block = (row // 100) == (ci // 100)
+    # ~90% locality + self edges + 10% random
+    assert same_block.mean() > 0.85
+    # and the 2-way partition halo becomes sparse -> halo comm mode
+    from roc_amd.parallel.partition import build_shard
+    sh = build_shard(g, 0, 2)
+    assert sh.halo_fraction < 0.5 and sh.comm_mode == "halo", \
+        (sh.halo_fraction, sh.comm_mode)
+    g2 = synthetic_graph(1000, 30000, seed=2)  # uniform control
+    sh2 = build_shard(g2, 0, 2)
+    assert sh2.comm_mode == "allgather"
