@@ -112,3 +112,35 @@ def test_auto_recover_cli(tmp_path):
         timeout=300)
     assert r2.returncode == 0, (r2.stdout[-800:], r2.stderr[-800:])
     assert "[recover]" in r2.stdout, r2.stdout[-800:]
+
+
+def test_train_from_reference_format_files(tmp_path):
+    """Full E2E on the reference's on-disk formats: .lux + feats.csv +
+    .label + .mask -> train.py --file."""
+    import numpy as np
+    from roc_amd.graph import synthetic_graph, save_lux
+    n, d, c = 120, 12, 5
+    g = synthetic_graph(n, 900, seed=21)
+    pref = str(tmp_path / "tiny")
+    save_lux(pref + ".add_self_edge.lux", g)
+    rng = np.random.default_rng(0)
+    np.savetxt(pref + ".feats.csv",
+               rng.standard_normal((n, d)).astype(np.float32), delimiter=",")
+    np.savetxt(pref + ".label", rng.integers(0, c, n), fmt="%d")
+    masks = rng.choice(["Train", "Val", "Test", "None"], size=n,
+                       p=[0.7, 0.1, 0.1, 0.1])
+    with open(pref + ".mask", "w") as f:
+        f.write("\n".join(masks) + "\n")
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(
+        os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "train.py", "--file", pref,
+         "--layers", f"{d}-8-{c}", "--epochs", "10", "--eval-every", "5"],
+        cwd=env["PYTHONPATH"], env=env, capture_output=True, text=True,
+        timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "epoch    10" in r.stdout, r.stdout
+    # binary feature cache was written on first load (load_task.cu:63-65)
+    assert os.path.exists(pref + ".feats.bin")
