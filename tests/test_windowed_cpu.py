@@ -86,3 +86,31 @@ def test_comm_plan_matches_scan_plan(tmp_path):
         assert err is None, f"rank {rank}: {err}"
         bad = [k for k, v in same.items() if not v]
         assert not bad, f"rank {rank}: mismatched {bad}"
+
+
+def test_train_from_lux_two_ranks(tmp_path):
+    """torchrun ws=2 over reference-format files: each rank reads only
+    its window (build_shard_from_lux + load_features_window)."""
+    import subprocess
+    import sys as _sys
+    import numpy as np
+    n, d, c = 160, 10, 4
+    g = synthetic_graph(n, 1300, seed=31)
+    pref = str(tmp_path / "w")
+    save_lux(pref + ".add_self_edge.lux", g)
+    rng = np.random.default_rng(1)
+    np.savetxt(pref + ".feats.csv",
+               rng.standard_normal((n, d)).astype(np.float32), delimiter=",")
+    np.savetxt(pref + ".label", rng.integers(0, c, n), fmt="%d")
+    with open(pref + ".mask", "w") as f:
+        f.write("\n".join(rng.choice(["Train", "Val", "Test"], size=n)) + "\n")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, PYTHONPATH=repo)
+    r = subprocess.run(
+        [_sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29621", "train.py", "--file", pref,
+         "--layers", f"{d}-8-{c}", "--epochs", "6", "--eval-every", "3"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1200:])
+    assert "epoch     6" in r.stdout, r.stdout[-800:]

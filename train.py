@@ -188,10 +188,13 @@ def main():
             print(f"resumed from {args.resume} at epoch {trainer.epoch}")
 
     if rank == 0:
-        print(f"[config] model={args.model} dims={dims} nodes={g.num_nodes} "
-              f"edges={g.num_edges} world={world} device={device} "
+        nn_ = g.num_nodes if g is not None else shard.bounds[-1]
+        ne_ = (g.num_edges if g is not None
+               else f"window:{shard.colidx.numel()}")
+        print(f"[config] model={args.model} dims={dims} nodes={nn_} "
+              f"edges={ne_} world={world} device={device} "
               f"dtype={dtype} lr={args.lr} wd={args.weight_decay} "
-              f"dropout={args.dropout}", flush=True)
+              f"dropout={args.dropout} comm={shard.comm_mode}", flush=True)
 
     if args.rebalance_every:
         trainer.attach_full_graph(g)
