@@ -83,3 +83,17 @@ def test_community_locality():
     g2 = synthetic_graph(1000, 30000, seed=2)  # dense uniform control
     sh2 = build_shard(g2, 0, 2)
     assert sh2.comm_mode == "allgather"
+
+
+def test_reorder_graph_preserves_structure():
+    from roc_amd.graph import reorder_graph, degree_order
+    from roc_amd.parallel.partition import build_shard
+    from roc_amd.ops import functional as F
+    g = synthetic_graph(80, 600, seed=9)
+    perm = degree_order(g)
+    g2 = reorder_graph(g, perm)
+    x = torch.randn(80, 6)
+    out1 = F.scatter_gather(x, build_shard(g, 0, 1))
+    out2 = F.scatter_gather(x[perm], build_shard(g2, 0, 1))
+    # aggregation commutes with relabeling
+    assert torch.allclose(out2, out1[perm], atol=1e-5)
