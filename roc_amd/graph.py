@@ -232,18 +232,17 @@ def reorder_graph(g: CSRGraph, perm: torch.Tensor) -> CSRGraph:
     deg = np.diff(rp)[perm_np]
     new_rp = np.zeros(g.num_nodes + 1, dtype=np.int64)
     np.cumsum(deg, out=new_rp[1:])
-    new_ci = np.empty(g.num_edges, dtype=np.int32)
-    for i0 in range(0, g.num_nodes, 1 << 20):   # chunked copy
-        i1 = min(g.num_nodes, i0 + (1 << 20))
-        # gather each new row's old edge span
-        for new_v in range(i0, i1):
-            old_v = perm_np[new_v]
-            s0, s1 = rp[old_v], rp[old_v + 1]
-            d0 = new_rp[new_v]
-            new_ci[d0:d0 + (s1 - s0)] = inv[g.colidx.numpy()[s0:s1]]
+    # vectorized edge permutation: new edge i maps to
+    # rp[perm[row(i)]] + (i - new_rp[row(i)])
+    row_of_new = np.repeat(np.arange(g.num_nodes, dtype=np.int64), deg)
+    old_edge = (np.repeat(rp[perm_np], deg)
+                + np.arange(g.num_edges, dtype=np.int64)
+                - np.repeat(new_rp[:-1], deg))
+    new_ci = inv[g.colidx.numpy()[old_edge]].astype(np.int32)
+    del row_of_new
     out = CSRGraph(num_nodes=g.num_nodes, num_edges=g.num_edges,
                    rowptr=torch.from_numpy(new_rp),
-                   colidx=torch.from_numpy(new_ci))
+                   colidx=torch.from_numpy(np.ascontiguousarray(new_ci)))
     try:
         from roc_amd import _C
         _C.csr_sort_rows(out.rowptr, out.colidx)
