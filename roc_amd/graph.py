@@ -234,12 +234,10 @@ def reorder_graph(g: CSRGraph, perm: torch.Tensor) -> CSRGraph:
     np.cumsum(deg, out=new_rp[1:])
     # vectorized edge permutation: new edge i maps to
     # rp[perm[row(i)]] + (i - new_rp[row(i)])
-    row_of_new = np.repeat(np.arange(g.num_nodes, dtype=np.int64), deg)
     old_edge = (np.repeat(rp[perm_np], deg)
                 + np.arange(g.num_edges, dtype=np.int64)
                 - np.repeat(new_rp[:-1], deg))
     new_ci = inv[g.colidx.numpy()[old_edge]].astype(np.int32)
-    del row_of_new
     out = CSRGraph(num_nodes=g.num_nodes, num_edges=g.num_edges,
                    rowptr=torch.from_numpy(new_rp),
                    colidx=torch.from_numpy(np.ascontiguousarray(new_ci)))
