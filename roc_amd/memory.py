@@ -40,11 +40,15 @@ class ActivationOffload:
         self._entries = []  # live offloaded entries, forward order
 
     # -- pinned buffer pool -------------------------------------------------
-    def _acquire(self, t: torch.Tensor) -> torch.Tensor:
+    def _acquire(self, t: torch.Tensor):
         key = (tuple(t.shape), t.dtype)
         free = self._pool.get(key)
         if free:
             return free.pop()
+        nbytes = t.numel() * t.element_size()
+        if self._pinned_total + nbytes > self.max_pinned_bytes:
+            return None  # budget reached: caller keeps the tensor resident
+        self._pinned_total += nbytes
         return torch.empty(t.shape, dtype=t.dtype, device="cpu",
                            pin_memory=True)
 
@@ -61,6 +65,8 @@ class ActivationOffload:
             return t
         cur = torch.cuda.current_stream(t.device)
         cpu = self._acquire(t)
+        if cpu is None:
+            return t  # pinned budget exhausted: stay HBM-resident
         self.stream.wait_stream(cur)          # producer finished
         with torch.cuda.stream(self.stream):
             cpu.copy_(t, non_blocking=True)
