@@ -33,6 +33,14 @@ class ActivationOffload:
         if os.environ.get("ROC_OFFLOAD_PREFETCH", "1") == "0":
             prefetch_window = 0
         self.prefetch_window = prefetch_window
+        # Hard budget on page-locked host memory: beyond it tensors stay
+        # HBM-resident (graceful degradation). Unbounded pinning can
+        # destabilize the HOST (hipHostMalloc is non-swappable; a 200+GB
+        # pinned footprint took a box down). ROC_OFFLOAD_MAX_GB overrides.
+        env_gb = os.environ.get("ROC_OFFLOAD_MAX_GB")
+        self.max_pinned_bytes = (int(float(env_gb) * (1 << 30)) if env_gb
+                                 else (64 << 30))
+        self._pinned_total = 0
         self.stream = torch.cuda.Stream() if self.enabled else None
         self._pool = {}   # (shape, dtype) -> list of free pinned tensors
         self._stats = {"offloaded_bytes": 0, "tensors": 0}
