@@ -21,9 +21,7 @@ import time
 import torch
 import torch.distributed as dist
 
-from roc_amd import (synthetic_dataset, build_shard, build_model,
-                     AdamOptimizer, Trainer)
-from roc_amd.parallel.partition import edge_balanced_bounds
+from roc_amd import synthetic_dataset, build_model, AdamOptimizer, Trainer
 
 
 def pad_features(feats: torch.Tensor, mult: int = 8) -> torch.Tensor:
