@@ -144,3 +144,18 @@ def test_train_from_reference_format_files(tmp_path):
     assert "epoch    10" in r.stdout, r.stdout
     # binary feature cache was written on first load (load_task.cu:63-65)
     assert os.path.exists(pref + ".feats.bin")
+
+
+def test_all_config_yamls_run(tmp_path):
+    """Every shipped BASELINE config YAML parses and drives a (tiny) run."""
+    import glob
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, PYTHONPATH=repo)
+    for cfg in sorted(glob.glob(os.path.join(repo, "configs", "*.yaml"))):
+        r = subprocess.run(
+            [sys.executable, "train.py", "--config", cfg,
+             "--scale", "0.002", "--epochs", "2", "--eval-every", "0",
+             "--hidden", "8"],
+            cwd=repo, env=env, capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, (cfg, r.stderr[-1500:])
+        assert "[config]" in r.stdout, (cfg, r.stdout[-500:])
