@@ -104,3 +104,18 @@ def test_capture_trains():
     m1 = tr.evaluate()
     F.set_dropout_counter(None)
     assert m1["ce_loss"] < m0["ce_loss"], (m0, m1)
+
+
+def test_capture_replay_no_leak():
+    """200 graph replays must not grow device memory (stable pool)."""
+    tr = _make(True, dropout=0.3)
+    for _ in range(10):
+        tr.train_epoch()
+    torch.cuda.synchronize()
+    base = torch.cuda.memory_allocated()
+    for _ in range(200):
+        tr.train_epoch()
+    torch.cuda.synchronize()
+    grown = torch.cuda.memory_allocated() - base
+    F.set_dropout_counter(None)
+    assert grown < (32 << 20), f"memory grew {grown/1e6:.1f} MB over replays"

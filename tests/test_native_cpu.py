@@ -59,3 +59,17 @@ def test_divergence_guard():
     check_metrics(torch.ones(8))
     with pytest.raises(TrainingDiverged):
         check_metrics(torch.tensor([1.0, float("nan"), 0, 0, 0, 0, 0, 0]))
+
+
+def test_sync_debug_mode():
+    from roc_amd.debug import (enable_sync_debug, check_tensor,
+                               sync_debug_enabled, TrainingDiverged)
+    enable_sync_debug(True)
+    try:
+        assert sync_debug_enabled()
+        check_tensor(torch.ones(4), "ok-tensor")
+        with pytest.raises(TrainingDiverged):
+            check_tensor(torch.tensor([1.0, float("inf")]), "bad-tensor")
+    finally:
+        enable_sync_debug(False)
+    assert not sync_debug_enabled()
