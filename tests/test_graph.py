@@ -1,4 +1,5 @@
 import numpy as np
+import pytest
 import torch
 
 from roc_amd.graph import (CSRGraph, load_lux, save_lux, synthetic_graph,
@@ -83,6 +84,28 @@ def test_community_locality():
     g2 = synthetic_graph(1000, 30000, seed=2)  # dense uniform control
     sh2 = build_shard(g2, 0, 2)
     assert sh2.comm_mode == "allgather"
+
+
+def test_label_and_mask_text_loaders(tmp_path):
+    """Reference on-disk formats: `.label` one class id per line
+    (`load_task.cu:110-123`), `.mask` Train|Val|Test|None strings
+    (`load_task.cu:160-183`)."""
+    from roc_amd.graph import (load_labels, load_mask, MASK_TRAIN, MASK_VAL,
+                               MASK_TEST, MASK_NONE)
+    lp = tmp_path / "d.label"
+    lp.write_text("3\n0\n7\n1\n")
+    labels = load_labels(str(lp), 4)
+    assert labels.dtype == torch.int64
+    assert labels.tolist() == [3, 0, 7, 1]
+    mp = tmp_path / "d.mask"
+    mp.write_text("Train\nVal\nTest\nNone\n")
+    mask = load_mask(str(mp), 4)
+    assert mask.tolist() == [MASK_TRAIN, MASK_VAL, MASK_TEST, MASK_NONE]
+    # row-count mismatch must fail loudly, not truncate
+    with pytest.raises(AssertionError):
+        load_labels(str(lp), 5)
+    with pytest.raises(AssertionError):
+        load_mask(str(mp), 3)
 
 
 def test_reorder_graph_preserves_structure():
