@@ -14,6 +14,7 @@ Reference flag parity: --file (.lux dataset prefix), --layers D0-D1-...-C,
 """
 import argparse
 import os
+import sys
 import time
 
 import torch
@@ -149,6 +150,11 @@ def main():
     if args.layers:
         dims = [int(d) for d in args.layers.split("-")]
         dims[0] = feats.shape[1]
+        if len(dims) < 2 or dims[-1] < num_classes:
+            sys.exit(f"--layers must give >=2 dash-separated dims ending "
+                     f">= num_classes={num_classes}, e.g. 602-256-41 "
+                     f"(got {args.layers!r}; for a layer COUNT use "
+                     f"--num-layers N)")
     else:
         c_out = num_classes
         if on_gpu:  # pad class dim to whole 128-B logits rows (see bench.py)
