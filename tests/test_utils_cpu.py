@@ -87,6 +87,21 @@ def test_train_cli_runs(tmp_path):
     assert "resumed" in r2.stdout
 
 
+def test_train_cli_rejects_bad_layers():
+    """--layers is the reference's dash-separated dims string
+    (`gnn.cc:130-143`); a bare count must fail with a clear message,
+    not a shape error deep in the kernels."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "train.py", "--dataset", "cora-synthetic",
+         "--scale", "0.04", "--layers", "4", "--epochs", "1"],
+        cwd=env["PYTHONPATH"], env=env, capture_output=True, text=True,
+        timeout=300)
+    assert r.returncode != 0
+    assert "--num-layers" in (r.stderr + r.stdout)
+
+
 def test_auto_recover_cli(tmp_path):
     """Divergence guard + checkpoint restore: train with an absurd LR that
     NaNs out; --auto-recover must restore and finish."""
