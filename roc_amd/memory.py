@@ -17,6 +17,8 @@ from __future__ import annotations
 
 import torch
 
+from . import streamcheck
+
 
 class ActivationOffload:
     """Context manager: offload saved activations >= min_bytes to host.
@@ -80,6 +82,7 @@ class ActivationOffload:
             cpu.copy_(t, non_blocking=True)
             ev = torch.cuda.Event()
             ev.record(self.stream)
+        streamcheck.producer(ev, "offload-d2h")
         t.record_stream(self.stream)          # allocator: defer reuse
         self._stats["offloaded_bytes"] += t.numel() * t.element_size()
         self._stats["tensors"] += 1
@@ -114,12 +117,14 @@ class ActivationOffload:
                 return
 
     def _start_h2d(self, entry):
+        streamcheck.consumer(entry["ev"], "offload-h2d")
         with torch.cuda.stream(self.stream):
             entry["ev"].wait(self.stream)     # D2H done before H2D
             entry["gpu"] = entry["cpu"].to(entry["device"], non_blocking=True)
             ev2 = torch.cuda.Event()
             ev2.record(self.stream)
             entry["ev2"] = ev2
+        streamcheck.producer(ev2, "offload-h2d")
 
     def _unpack(self, packed):
         if not isinstance(packed, tuple) or not packed or \
@@ -132,6 +137,7 @@ class ActivationOffload:
                 "not supported with ActivationOffload)")
         if entry["gpu"] is None:
             self._start_h2d(entry)
+        streamcheck.consumer(entry["ev2"], "backward-consume")
         cur = torch.cuda.current_stream(entry["device"])
         cur.wait_event(entry["ev2"])
         gpu = entry["gpu"]

@@ -14,6 +14,8 @@ from __future__ import annotations
 import torch
 import torch.distributed as dist
 
+from .. import streamcheck
+
 
 def _a2a(out: torch.Tensor, inp: torch.Tensor, out_splits, in_splits, group):
     dist.all_to_all_single(
@@ -109,10 +111,12 @@ class _HaloAggregate(torch.autograd.Function):
         work = dist.all_to_all_single(
             halo, send, output_split_sizes=shard.recv_splits,
             input_split_sizes=shard.send_splits, group=group, async_op=True)
+        streamcheck.producer(work, "halo-a2a-fwd")
         # 2) interior aggregation overlaps the exchange
         out = torch.empty(shard.n_local, D, dtype=x.dtype, device=x.device)
         _spmm_part(out, x, shard.loc_rowptr, shard.loc_colidx, None, False)
         # 3) boundary contribution + final dst scaling
+        streamcheck.consumer(work, "boundary-spmm")
         work.wait()
         _spmm_part(out, halo, shard.halo_rowptr, shard.halo_colidx,
                    dst_scale, True)
@@ -136,10 +140,12 @@ class _HaloAggregate(torch.autograd.Function):
         work = dist.all_to_all_single(
             grad_in, dhalo, output_split_sizes=shard.send_splits,
             input_split_sizes=shard.recv_splits, group=group, async_op=True)
+        streamcheck.producer(work, "halo-a2a-bwd")
         # 2) local-source grads overlap the exchange
         dx = torch.empty(shard.n_local, D, dtype=dy.dtype, device=dy.device)
         _spmm_part(dx, dy, shard.t_loc_rowptr, shard.t_loc_colidx,
                    None, False)
+        streamcheck.consumer(work, "grad-scatter")
         work.wait()
         dx.index_add_(0, shard.send_idx, grad_in)
         return dx, None, None, None

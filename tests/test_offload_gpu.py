@@ -49,3 +49,26 @@ def test_offload_stats_grow():
     torch.cuda.synchronize()
     s = tr.offload.stats
     assert s["offloaded_bytes"] > 0
+
+
+def test_offload_declares_stream_edges():
+    """Happens-before debug mode (SURVEY §5 race detection): every D2H
+    and H2D copy edge in the offload tier is declared and validated —
+    d2h -> h2d -> backward-consume per offloaded tensor, none leaked."""
+    from roc_amd import streamcheck as sc
+    tr = _make(offload=True)
+    sc.enable_stream_debug(True)
+    try:
+        tr.train_epoch()
+        torch.cuda.synchronize()
+        edges = sc.edge_log()
+        n = tr.offload.stats["tensors"]
+        assert n > 0
+        h2d = [e for e in edges if e == ("offload-d2h", "offload-h2d")]
+        consume = [e for e in edges if e == ("offload-h2d",
+                                             "backward-consume")]
+        assert len(h2d) == n, (n, edges)
+        assert len(consume) == n, (n, edges)
+        assert sc.pending() == []
+    finally:
+        sc.enable_stream_debug(False)
