@@ -20,7 +20,16 @@ from __future__ import annotations
 import torch
 import torch.distributed as dist
 
-from .halo import halo_exchange, halo_aggregate, overlap_enabled
+import os as _os
+
+from .. import streamcheck
+from .halo import halo_exchange, halo_aggregate, overlap_enabled, _spmm_part
+
+
+def ag_overlap_enabled() -> bool:
+    """Overlap the forward all_gather with the self-block SpMM
+    (default ON; ROC_AG_OVERLAP=0 falls back to the sequential path)."""
+    return _os.environ.get("ROC_AG_OVERLAP", "1") == "1"
 
 
 def _reduce_scatter(out, inp, group):
@@ -63,10 +72,25 @@ class _GatherAggregate(torch.autograd.Function):
             send = torch.zeros(mr, D, dtype=x.dtype, device=x.device)
             send[:x.shape[0]] = x
         gathered = torch.empty(ws * mr, D, dtype=x.dtype, device=x.device)
-        dist.all_gather_into_tensor(gathered, send.contiguous(), group=group)
         out = torch.empty(shard.n_local, D, dtype=x.dtype, device=x.device)
-        _spmm_local(out, gathered, shard.rowptr, shard.ag_colidx, dst_scale,
-                    shard.row_order if out.is_cuda else None)
+        if ag_overlap_enabled() and shard.ag_self_rowptr is not None:
+            # self-source edges aggregate from local x while the gather
+            # is in flight; remote-source edges then accumulate and the
+            # dst scale lands on the final store
+            work = dist.all_gather_into_tensor(
+                gathered, send.contiguous(), group=group, async_op=True)
+            streamcheck.producer(work, "ag-allgather-fwd")
+            _spmm_part(out, x, shard.ag_self_rowptr, shard.ag_self_colidx,
+                       None, False)
+            streamcheck.consumer(work, "ag-remote-accumulate")
+            work.wait()
+            _spmm_part(out, gathered, shard.ag_rem_rowptr,
+                       shard.ag_rem_colidx, dst_scale, True)
+        else:
+            dist.all_gather_into_tensor(gathered, send.contiguous(),
+                                        group=group)
+            _spmm_local(out, gathered, shard.rowptr, shard.ag_colidx,
+                        dst_scale, shard.row_order if out.is_cuda else None)
         return out
 
     @staticmethod

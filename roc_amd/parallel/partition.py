@@ -69,6 +69,14 @@ class GraphShard:
     ag_t_rowptr: Optional[torch.Tensor] = None   # [ws*ag_max_rows+1]
     ag_t_colidx: Optional[torch.Tensor] = None
     ag_t_row_order: Optional[torch.Tensor] = None
+    # self/remote split of the gather-space CSR: edges whose source this
+    # rank owns aggregate straight from local x while the all_gather is
+    # in flight, then remote-source edges accumulate from the gathered
+    # buffer (the allgather-mode analogue of the halo loc/halo split)
+    ag_self_rowptr: Optional[torch.Tensor] = None  # [n_local+1], cols<n_local
+    ag_self_colidx: Optional[torch.Tensor] = None
+    ag_rem_rowptr: Optional[torch.Tensor] = None   # [n_local+1], gather space
+    ag_rem_colidx: Optional[torch.Tensor] = None
 
     def to(self, device) -> "GraphShard":
         d = {}
@@ -267,6 +275,21 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
         ag_t_deg = np.diff(ag_t[0].numpy())
         shard_kw["ag_t_row_order"] = torch.from_numpy(
             np.argsort(-ag_t_deg, kind="stable").astype(np.int32))
+        # self/remote edge split for comm/compute overlap (fwd only)
+        is_self = owners == rank
+        row_of_edge = np.repeat(np.arange(n_local, dtype=np.int64),
+                                np.diff(local_rowptr))
+        for key_pfx, sel, cols in (
+                ("ag_self", is_self,
+                 (local_cols_global - lo).astype(np.int32)),
+                ("ag_rem", ~is_self, ag_cols.astype(np.int32))):
+            rows_sel = row_of_edge[sel]
+            cnt = np.bincount(rows_sel, minlength=n_local).astype(np.int64)
+            part_rp = np.zeros(n_local + 1, dtype=np.int64)
+            np.cumsum(cnt, out=part_rp[1:])
+            shard_kw[f"{key_pfx}_rowptr"] = torch.from_numpy(part_rp)
+            shard_kw[f"{key_pfx}_colidx"] = torch.from_numpy(
+                np.ascontiguousarray(cols[sel]))
     if world_size > 1 and comm_mode == "halo":
         # split edges by source locality for comm/compute overlap
         is_loc_edge = colidx < n_local

@@ -64,6 +64,62 @@ def _worker(rank, port, model_name, q):
             dist.destroy_process_group()
 
 
+def _ab_worker(rank, port, q):
+    """Allgather overlap (async gather + self/remote SpMM split) vs the
+    sequential gather->SpMM path: identical weights after 3 epochs."""
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["ROC_COMM_MODE"] = "allgather"
+        dist.init_process_group("gloo", rank=rank, world_size=WS)
+        g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.08,
+                                                      seed=13)
+        bounds = edge_balanced_bounds(g.rowptr, WS)
+        sh = build_shard(g, rank, WS, bounds)
+        assert sh.ag_self_rowptr is not None
+        # split covers every edge exactly once
+        n_split = sh.ag_self_colidx.numel() + sh.ag_rem_colidx.numel()
+        assert n_split == sh.ag_colidx.numel()
+        ws = {}
+        for ov in ("1", "0"):
+            os.environ["ROC_AG_OVERLAP"] = ov
+            model = build_model("gcn", [feats.shape[1], 16, c],
+                                dropout=0.0, seed=1)
+            opt = AdamOptimizer(model.parameters(), lr=0.01)
+            tr = Trainer(model, sh, feats, labels, mask, opt)
+            for _ in range(3):
+                tr.train_epoch()
+            p0 = next(iter(model.parameters()))
+            ws[ov] = p0.detach().numpy().copy()
+        q.put((rank, ws, None))
+    except Exception:  # pragma: no cover
+        import traceback
+        q.put((rank, None, traceback.format_exc()))
+    finally:
+        os.environ.pop("ROC_COMM_MODE", None)
+        os.environ.pop("ROC_AG_OVERLAP", None)
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_ag_overlap_matches_sequential():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_ab_worker, args=(r, 29597, q))
+             for r in range(WS)]
+    for p in procs:
+        p.start()
+    res = sorted([q.get() for _ in range(WS)], key=lambda t: t[0])
+    for p in procs:
+        p.join(timeout=180)
+    for rank, ws, err in res:
+        assert err is None, f"rank {rank}: {err}"
+        a = torch.from_numpy(ws["1"])
+        b = torch.from_numpy(ws["0"])
+        assert torch.allclose(a, b, atol=1e-5), \
+            (rank, (a - b).abs().max())
+
+
 @pytest.mark.parametrize("model_name,port", [("gcn", 29591), ("sage", 29593),
                                              ("gin", 29595)])
 def test_allgather_matches_halo(model_name, port):
