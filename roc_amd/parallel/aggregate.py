@@ -27,9 +27,23 @@ from .halo import halo_exchange, halo_aggregate, overlap_enabled, _spmm_part
 
 
 def ag_overlap_enabled() -> bool:
-    """Overlap the forward all_gather with the self-block SpMM
-    (default ON; ROC_AG_OVERLAP=0 falls back to the sequential path)."""
+    """Overlap the allgather-mode collectives with the aggregation SpMMs
+    (default ON; ROC_AG_OVERLAP=0 falls back to the sequential
+    all_gather -> SpMM forward and SpMM^T -> reduce_scatter backward)."""
     return _os.environ.get("ROC_AG_OVERLAP", "1") == "1"
+
+
+def _ag_block_bounds(shard):
+    """Edge offsets of each owner block in ag_t_colidx ([ws+1] python
+    ints, cached on the shard: one D2H sync total, not one per layer)."""
+    b = getattr(shard, "_ag_t_bounds", None)
+    if b is None:
+        mr = shard.ag_max_rows
+        idx = torch.arange(0, shard.world_size * mr + 1, mr,
+                           device=shard.ag_t_rowptr.device)
+        b = [int(v) for v in shard.ag_t_rowptr[idx].cpu()]
+        object.__setattr__(shard, "_ag_t_bounds", b)
+    return b
 
 
 def _reduce_scatter(out, inp, group):
@@ -105,6 +119,28 @@ class _GatherAggregate(torch.autograd.Function):
         mr = shard.ag_max_rows
         ws = shard.world_size
         dfull = torch.empty(ws * mr, D, dtype=dy.dtype, device=dy.device)
+        if ag_overlap_enabled():
+            # per-owner-block SpMM^T + async reduce-to-owner: block r's
+            # reduction is in flight while block r+1 computes (same total
+            # bytes as reduce_scatter, comm hidden behind compute)
+            bnds = _ag_block_bounds(shard)
+            works = []
+            for r in range(ws):  # same launch order on every rank
+                rp_blk = shard.ag_t_rowptr[r * mr:(r + 1) * mr + 1]
+                rp_blk = (rp_blk - rp_blk[:1]).contiguous()
+                cols = shard.ag_t_colidx[bnds[r]:bnds[r + 1]]
+                blk = dfull[r * mr:(r + 1) * mr]
+                _spmm_part(blk, dy, rp_blk, cols, None, False)
+                dst = (dist.get_global_rank(group, r)
+                       if group is not None else r)
+                w = dist.reduce(blk, dst=dst, group=group, async_op=True)
+                streamcheck.producer(w, f"ag-reduce-bwd")
+                works.append(w)
+            for w in works:
+                streamcheck.consumer(w, "ag-bwd-done")
+                w.wait()
+            own = dfull[shard.rank * mr:shard.rank * mr + shard.n_local]
+            return own.contiguous(), None, None, None
         _spmm_local(dfull, dy, shard.ag_t_rowptr, shard.ag_t_colidx, None,
                     shard.ag_t_row_order if dy.is_cuda else None)
         dx_pad = torch.empty(mr, D, dtype=dy.dtype, device=dy.device)

@@ -81,6 +81,8 @@ class GraphShard:
     def to(self, device) -> "GraphShard":
         d = {}
         for k, v in self.__dict__.items():
+            if k.startswith("_"):  # lazy caches don't survive the move
+                continue
             d[k] = v.to(device) if isinstance(v, torch.Tensor) else v
         return GraphShard(**d)
 
