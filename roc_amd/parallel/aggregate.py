@@ -94,12 +94,15 @@ class _GatherAggregate(torch.autograd.Function):
             work = dist.all_gather_into_tensor(
                 gathered, send.contiguous(), group=group, async_op=True)
             streamcheck.producer(work, "ag-allgather-fwd")
+            cuda = out.is_cuda
             _spmm_part(out, x, shard.ag_self_rowptr, shard.ag_self_colidx,
-                       None, False)
+                       None, False,
+                       shard.ag_self_row_order if cuda else None)
             streamcheck.consumer(work, "ag-remote-accumulate")
             work.wait()
             _spmm_part(out, gathered, shard.ag_rem_rowptr,
-                       shard.ag_rem_colidx, dst_scale, True)
+                       shard.ag_rem_colidx, dst_scale, True,
+                       shard.ag_rem_row_order if cuda else None)
         else:
             dist.all_gather_into_tensor(gathered, send.contiguous(),
                                         group=group)
@@ -130,7 +133,10 @@ class _GatherAggregate(torch.autograd.Function):
                 rp_blk = (rp_blk - rp_blk[:1]).contiguous()
                 cols = shard.ag_t_colidx[bnds[r]:bnds[r + 1]]
                 blk = dfull[r * mr:(r + 1) * mr]
-                _spmm_part(blk, dy, rp_blk, cols, None, False)
+                order = (shard.ag_t_blk_order[r * mr:(r + 1) * mr]
+                         if dy.is_cuda and shard.ag_t_blk_order is not None
+                         else None)
+                _spmm_part(blk, dy, rp_blk, cols, None, False, order)
                 dst = (dist.get_global_rank(group, r)
                        if group is not None else r)
                 w = dist.reduce(blk, dst=dst, group=group, async_op=True)

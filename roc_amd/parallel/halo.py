@@ -82,11 +82,11 @@ def overlap_enabled() -> bool:
     return _os.environ.get("ROC_OVERLAP", "0") == "1"
 
 
-def _spmm_part(out, x, rowptr, colidx, dst, acc):
+def _spmm_part(out, x, rowptr, colidx, dst, acc, order=None):
     from ..ops import functional as Fn
     if out.is_cuda:
         Fn._hip(out)
-        Fn._C.spmm(out, x, rowptr, colidx, dst, None, None, acc)
+        Fn._C.spmm(out, x, rowptr, colidx, dst, None, order, acc)
     else:
         from ..ops import reference as ref
         part = ref.spmm(x, rowptr, colidx, out.shape[0])

@@ -77,6 +77,9 @@ class GraphShard:
     ag_self_colidx: Optional[torch.Tensor] = None
     ag_rem_rowptr: Optional[torch.Tensor] = None   # [n_local+1], gather space
     ag_rem_colidx: Optional[torch.Tensor] = None
+    ag_self_row_order: Optional[torch.Tensor] = None  # degree-desc, int32
+    ag_rem_row_order: Optional[torch.Tensor] = None
+    ag_t_blk_order: Optional[torch.Tensor] = None  # [ws*mr] block-local
 
     def to(self, device) -> "GraphShard":
         d = {}
@@ -292,6 +295,16 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
             shard_kw[f"{key_pfx}_rowptr"] = torch.from_numpy(part_rp)
             shard_kw[f"{key_pfx}_colidx"] = torch.from_numpy(
                 np.ascontiguousarray(cols[sel]))
+            shard_kw[f"{key_pfx}_row_order"] = torch.from_numpy(
+                np.argsort(-cnt, kind="stable").astype(np.int32))
+        # per-owner-block degree-desc row orders for the backward blocks
+        ag_t_rp = ag_t[0].numpy()
+        blk_ord = np.empty(world_size * mr, dtype=np.int32)
+        for r in range(world_size):
+            deg_blk = np.diff(ag_t_rp[r * mr:(r + 1) * mr + 1])
+            blk_ord[r * mr:(r + 1) * mr] = np.argsort(
+                -deg_blk, kind="stable").astype(np.int32)
+        shard_kw["ag_t_blk_order"] = torch.from_numpy(blk_ord)
     if world_size > 1 and comm_mode == "halo":
         # split edges by source locality for comm/compute overlap
         is_loc_edge = colidx < n_local

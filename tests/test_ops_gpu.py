@@ -351,3 +351,50 @@ def test_spmm_shape_sweep(graph_small):
                                              else 1e-5) + 1e-3
             assert torch.allclose(got, want, atol=tol, rtol=0.05), \
                 (D, dt, (got - want).abs().max())
+
+
+def test_ag_split_spmm_blocks_match_reference():
+    """The allgather-overlap kernel shapes: self/remote CSR splits with
+    degree-desc row orders, and per-owner-block transpose slices with
+    block-local orders + accumulate — exactly what the multi-GPU
+    aggregate path launches (parallel/aggregate.py)."""
+    from roc_amd.parallel.halo import _spmm_part
+    from roc_amd.parallel.aggregate import _ag_block_bounds
+    g = synthetic_graph(800, 24000, seed=21)
+    sh = build_shard(g, 0, 2)          # rank 0's view of a 2-way split
+    assert sh.comm_mode == "allgather" and sh.ag_self_rowptr is not None
+    D, ws, mr = 64, sh.world_size, sh.ag_max_rows
+    torch.manual_seed(3)
+    x_local = torch.randn(sh.n_local, D)
+    x_full = torch.randn(g.num_nodes, D)
+    x_full[:sh.n_local] = x_local
+    # padded gather-space buffer [ws*mr, D]
+    gathered = torch.zeros(ws * mr, D)
+    for r in range(ws):
+        blo, bhi = sh.bounds[r], sh.bounds[r + 1]
+        gathered[r * mr:r * mr + (bhi - blo)] = x_full[blo:bhi]
+    shd = sh.to(DEV)
+    # forward: self pass (order) + remote accumulate (order)
+    out = torch.empty(sh.n_local, D, device=DEV)
+    _spmm_part(out, x_local.to(DEV), shd.ag_self_rowptr, shd.ag_self_colidx,
+               None, False, shd.ag_self_row_order)
+    _spmm_part(out, gathered.to(DEV), shd.ag_rem_rowptr, shd.ag_rem_colidx,
+               None, True, shd.ag_rem_row_order)
+    want = ref.spmm(x_full, g.rowptr, g.colidx, g.num_nodes)[:sh.n_local]
+    assert torch.allclose(out.cpu(), want, rtol=1e-4, atol=1e-3), \
+        (out.cpu() - want).abs().max()
+    # backward blocks: per-owner slice + block-local order
+    dy = torch.randn(sh.n_local, D)
+    bnds = _ag_block_bounds(shd)
+    dfull = torch.empty(ws * mr, D, device=DEV)
+    dyd = dy.to(DEV)
+    for r in range(ws):
+        rp_blk = shd.ag_t_rowptr[r * mr:(r + 1) * mr + 1]
+        rp_blk = (rp_blk - rp_blk[:1]).contiguous()
+        cols = shd.ag_t_colidx[bnds[r]:bnds[r + 1]]
+        order = shd.ag_t_blk_order[r * mr:(r + 1) * mr]
+        _spmm_part(dfull[r * mr:(r + 1) * mr], dyd, rp_blk, cols,
+                   None, False, order)
+    want_full = ref.spmm(dy, sh.ag_t_rowptr, sh.ag_t_colidx, ws * mr)
+    assert torch.allclose(dfull.cpu(), want_full, rtol=1e-4, atol=1e-3), \
+        (dfull.cpu() - want_full).abs().max()
