@@ -140,7 +140,7 @@ class _GatherAggregate(torch.autograd.Function):
                 dst = (dist.get_global_rank(group, r)
                        if group is not None else r)
                 w = dist.reduce(blk, dst=dst, group=group, async_op=True)
-                streamcheck.producer(w, f"ag-reduce-bwd")
+                streamcheck.producer(w, "ag-reduce-bwd")
                 works.append(w)
             for w in works:
                 streamcheck.consumer(w, "ag-bwd-done")
