@@ -45,7 +45,8 @@ class ActivationOffload:
         self._pinned_total = 0
         self.stream = torch.cuda.Stream() if self.enabled else None
         self._pool = {}   # (shape, dtype) -> list of free pinned tensors
-        self._stats = {"offloaded_bytes": 0, "tensors": 0}
+        self._stats = {"offloaded_bytes": 0, "tensors": 0,
+                       "budget_skipped": 0}
         self._ctx = None
         self._entries = []  # live offloaded entries, forward order
 
@@ -76,6 +77,7 @@ class ActivationOffload:
         cur = torch.cuda.current_stream(t.device)
         cpu = self._acquire(t)
         if cpu is None:
+            self._stats["budget_skipped"] += 1
             return t  # pinned budget exhausted: stay HBM-resident
         self.stream.wait_stream(cur)          # producer finished
         with torch.cuda.stream(self.stream):
