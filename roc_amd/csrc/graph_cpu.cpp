@@ -60,6 +60,138 @@ std::vector<torch::Tensor> csr_transpose(int64_t num_cols,
   return {t_rowptr, t_colidx};
 }
 
+// Reverse Cuthill-McKee ordering on the symmetrized adjacency (in-CSR +
+// its transpose, both passed in so the O(E) counting-sort transpose is
+// reused). Classic RCM: BFS from a minimum-degree seed per component,
+// neighbors enqueued degree-ascending, final order reversed. Returns an
+// int64 permutation for `reorder_graph` (a locality preprocessing the
+// reference lacks; its partitioner `gnn.cc:806-829` cuts contiguous
+// ranges of whatever order the dataset shipped).
+torch::Tensor rcm_order(torch::Tensor rowptr, torch::Tensor colidx,
+                        torch::Tensor t_rowptr, torch::Tensor t_colidx) {
+  TORCH_CHECK(rowptr.device().is_cpu() && colidx.device().is_cpu());
+  TORCH_CHECK(rowptr.scalar_type() == torch::kInt64);
+  TORCH_CHECK(colidx.scalar_type() == torch::kInt32);
+  rowptr = rowptr.contiguous();
+  colidx = colidx.contiguous();
+  t_rowptr = t_rowptr.contiguous();
+  t_colidx = t_colidx.contiguous();
+  const int64_t n = rowptr.numel() - 1;
+  TORCH_CHECK(t_rowptr.numel() - 1 == n, "transpose must be square");
+  const int64_t* rp = rowptr.data_ptr<int64_t>();
+  const int* ci = colidx.data_ptr<int>();
+  const int64_t* trp = t_rowptr.data_ptr<int64_t>();
+  const int* tci = t_colidx.data_ptr<int>();
+
+  std::vector<int64_t> deg(n);
+#pragma omp parallel for schedule(static)
+  for (int64_t v = 0; v < n; ++v)
+    deg[v] = (rp[v + 1] - rp[v]) + (trp[v + 1] - trp[v]);
+
+  // seeds: degree-ascending so each component starts at a pseudo-peripheral
+  // low-degree vertex (counting sort — degrees are bounded by 2E)
+  std::vector<int64_t> seeds(n);
+  {
+    std::vector<int64_t> idx(n);
+    for (int64_t v = 0; v < n; ++v) idx[v] = v;
+    std::stable_sort(idx.begin(), idx.end(),
+                     [&](int64_t a, int64_t b) { return deg[a] < deg[b]; });
+    seeds.swap(idx);
+  }
+
+  auto out = torch::empty({n}, torch::kInt64);
+  int64_t* order = out.data_ptr<int64_t>();  // doubles as the BFS queue
+  std::vector<uint8_t> visited(n, 0);
+  std::vector<int64_t> nbr;
+  int64_t head = 0, tail = 0;
+  size_t seed_i = 0;
+  while (tail < n) {
+    while (seed_i < seeds.size() && visited[seeds[seed_i]]) ++seed_i;
+    const int64_t s = seeds[seed_i];
+    visited[s] = 1;
+    order[tail++] = s;
+    while (head < tail) {
+      const int64_t v = order[head++];
+      nbr.clear();
+      for (int64_t e = rp[v]; e < rp[v + 1]; ++e) {
+        const int64_t u = ci[e];
+        if (!visited[u]) { visited[u] = 1; nbr.push_back(u); }
+      }
+      for (int64_t e = trp[v]; e < trp[v + 1]; ++e) {
+        const int64_t u = tci[e];
+        if (!visited[u]) { visited[u] = 1; nbr.push_back(u); }
+      }
+      std::sort(nbr.begin(), nbr.end(),
+                [&](int64_t a, int64_t b) {
+                  return deg[a] != deg[b] ? deg[a] < deg[b] : a < b;
+                });
+      for (const int64_t u : nbr) order[tail++] = u;
+    }
+  }
+  std::reverse(order, order + n);
+  return out;
+}
+
+// Label-propagation clustering order: T synchronous rounds where each
+// vertex adopts the most frequent label among its in-neighbors (the CSR
+// row, which includes the self-edge), then a stable sort by final
+// label. On community-structured graphs this packs each community
+// contiguously — far better gather locality than RCM, whose BFS fronts
+// explode through the long-range edges. In-edges-only measurably beats
+// symmetrized counting here (hub out-fans otherwise smear labels).
+// Deterministic: ties break toward the smaller label.
+torch::Tensor lp_cluster_order(torch::Tensor rowptr, torch::Tensor colidx,
+                               int64_t iters) {
+  TORCH_CHECK(rowptr.device().is_cpu() && colidx.device().is_cpu());
+  TORCH_CHECK(rowptr.scalar_type() == torch::kInt64);
+  TORCH_CHECK(colidx.scalar_type() == torch::kInt32);
+  rowptr = rowptr.contiguous();
+  colidx = colidx.contiguous();
+  const int64_t n = rowptr.numel() - 1;
+  const int64_t* rp = rowptr.data_ptr<int64_t>();
+  const int* ci = colidx.data_ptr<int>();
+
+  std::vector<int64_t> lab(n), next(n);
+  for (int64_t v = 0; v < n; ++v) lab[v] = v;
+  std::atomic<int64_t> changed{0};
+  for (int64_t it = 0; it < iters; ++it) {
+    changed.store(0, std::memory_order_relaxed);
+#pragma omp parallel
+    {
+      std::vector<int64_t> buf;
+#pragma omp for schedule(dynamic, 2048)
+      for (int64_t v = 0; v < n; ++v) {
+        buf.clear();
+        for (int64_t e = rp[v]; e < rp[v + 1]; ++e) buf.push_back(lab[ci[e]]);
+        if (buf.empty()) { next[v] = lab[v]; continue; }
+        std::sort(buf.begin(), buf.end());
+        int64_t best = buf[0], best_cnt = 1, cur = buf[0], cur_cnt = 1;
+        for (size_t i = 1; i < buf.size(); ++i) {
+          if (buf[i] == cur) {
+            ++cur_cnt;
+          } else {
+            if (cur_cnt > best_cnt) { best = cur; best_cnt = cur_cnt; }
+            cur = buf[i];
+            cur_cnt = 1;
+          }
+        }
+        if (cur_cnt > best_cnt) { best = cur; best_cnt = cur_cnt; }
+        next[v] = best;
+        if (best != lab[v]) changed.fetch_add(1, std::memory_order_relaxed);
+      }
+    }
+    lab.swap(next);
+    if (changed.load(std::memory_order_relaxed) == 0) break;
+  }
+
+  auto out = torch::empty({n}, torch::kInt64);
+  int64_t* order = out.data_ptr<int64_t>();
+  for (int64_t v = 0; v < n; ++v) order[v] = v;
+  std::stable_sort(order, order + n,
+                   [&](int64_t a, int64_t b) { return lab[a] < lab[b]; });
+  return out;
+}
+
 // sort column ids within each row (in place)
 void csr_sort_rows(torch::Tensor rowptr, torch::Tensor colidx) {
   TORCH_CHECK(rowptr.device().is_cpu() && colidx.device().is_cpu());
@@ -74,4 +206,8 @@ void register_graph_cpu(pybind11::module_& m) {
   m.def("csr_transpose", &csr_transpose,
         "counting-sort CSR transpose (CPU, OpenMP)");
   m.def("csr_sort_rows", &csr_sort_rows, "sort column ids within rows");
+  m.def("rcm_order", &rcm_order,
+        "reverse Cuthill-McKee permutation over CSR + transpose");
+  m.def("lp_cluster_order", &lp_cluster_order,
+        "label-propagation clustering permutation (OpenMP)");
 }

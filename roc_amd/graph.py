@@ -256,6 +256,56 @@ def degree_order(g: CSRGraph) -> torch.Tensor:
     return torch.from_numpy(np.argsort(-deg, kind="stable"))
 
 
+def rcm_order(g: CSRGraph) -> torch.Tensor:
+    """Reverse Cuthill-McKee relabeling permutation over the symmetrized
+    adjacency (in-edges + out-edges). Clusters each vertex next to its
+    neighbors, so (a) SpMM gathers hit an L2-sized source window and
+    (b) contiguous vertex-range partitions cut fewer edges (sparser
+    halos). Native BFS in ``_C.rcm_order`` (O(E) + per-front sort);
+    scipy fallback gives the same class of ordering."""
+    t_rowptr, t_colidx = build_transpose(g.num_nodes, g.rowptr, g.colidx)
+    try:
+        from roc_amd import _C
+        return _C.rcm_order(g.rowptr.contiguous(), g.colidx.contiguous(),
+                            t_rowptr.contiguous(), t_colidx.contiguous())
+    except ImportError:
+        from scipy.sparse import csr_matrix
+        from scipy.sparse.csgraph import reverse_cuthill_mckee
+        a = csr_matrix((np.ones(g.num_edges, dtype=np.int8),
+                        g.colidx.numpy(), g.rowptr.numpy()),
+                       shape=(g.num_nodes, g.num_nodes))
+        return torch.from_numpy(
+            reverse_cuthill_mckee(a, symmetric_mode=False).astype(np.int64))
+
+
+def cluster_order(g: CSRGraph, iters: int = 25) -> torch.Tensor:
+    """Label-propagation clustering relabeling: each vertex repeatedly
+    adopts its in-neighbors' majority label, then nodes are stably
+    sorted by final label — communities become contiguous id ranges.
+    The strongest of the three orderings on community graphs (see
+    profiles/): gathers land in an L2-sized window and contiguous
+    partitions cut few edges. Native OpenMP (``_C.lp_cluster_order``)."""
+    from roc_amd import _C
+    return _C.lp_cluster_order(g.rowptr.contiguous(), g.colidx.contiguous(),
+                               int(iters))
+
+
+ORDERINGS = {"degree": degree_order, "rcm": rcm_order,
+             "cluster": cluster_order}
+
+
+def apply_ordering(g: CSRGraph, feats: torch.Tensor, labels: torch.Tensor,
+                   mask: torch.Tensor, kind: str):
+    """Relabel the whole dataset by a named locality ordering ("degree" or
+    "rcm"); returns (graph, feats, labels, mask, perm). Deterministic, so
+    every rank computes the identical permutation. Training on the
+    relabeled dataset is mathematically the same full-graph problem
+    (permutation equivariance)."""
+    perm = ORDERINGS[kind](g)
+    return (reorder_graph(g, perm), feats[perm], labels[perm], mask[perm],
+            perm)
+
+
 # ---------------------------------------------------------------------------
 # Synthetic graphs (no-network benchmark datasets; BASELINE.json configs)
 # ---------------------------------------------------------------------------

@@ -120,3 +120,78 @@ def test_reorder_graph_preserves_structure():
     out2 = F.scatter_gather(x[perm], build_shard(g2, 0, 1))
     # aggregation commutes with relabeling
     assert torch.allclose(out2, out1[perm], atol=1e-5)
+
+
+def test_rcm_order_path_bandwidth():
+    # a label-shuffled path graph must come back to bandwidth <= 2
+    from roc_amd.graph import rcm_order, reorder_graph
+    n = 300
+    rng = np.random.default_rng(0)
+    shuf = rng.permutation(n)
+    rows = [[] for _ in range(n)]
+    for i in range(n - 1):
+        rows[shuf[i + 1]].append(int(shuf[i]))
+    rowptr = np.zeros(n + 1, dtype=np.int64)
+    cols = []
+    for v in range(n):
+        rowptr[v + 1] = rowptr[v] + len(rows[v])
+        cols.extend(sorted(rows[v]))
+    g = CSRGraph(num_nodes=n, num_edges=len(cols),
+                 rowptr=torch.from_numpy(rowptr),
+                 colidx=torch.tensor(cols, dtype=torch.int32))
+    perm = rcm_order(g)
+    assert sorted(perm.tolist()) == list(range(n))
+    g2 = reorder_graph(g, perm)
+    rp, ci = g2.rowptr.numpy(), g2.colidx.numpy()
+    r = np.repeat(np.arange(n), np.diff(rp))
+    assert np.abs(r - ci).max() <= 2
+
+
+def _window_frac(g, w=8192):
+    rp, ci = g.rowptr.numpy(), g.colidx.numpy()
+    rows = np.repeat(np.arange(g.num_nodes), np.diff(rp))
+    return float((np.abs(rows - ci) < w // 2).mean())
+
+
+def test_cluster_order_recovers_communities():
+    # shuffle a community graph; LP clustering must recover most of the
+    # gather locality the shuffle destroyed
+    from roc_amd.graph import cluster_order, reorder_graph
+    rng = np.random.default_rng(1)
+    n = 20000
+    g = synthetic_graph(n, 200_000, seed=3, locality=0.85,
+                        num_communities=20)
+    gshuf = reorder_graph(g, torch.from_numpy(rng.permutation(n)))
+    perm = cluster_order(gshuf)
+    assert sorted(perm.tolist()) == list(range(n))
+    glp = reorder_graph(gshuf, perm)
+    nat, shuf, rec = _window_frac(g), _window_frac(gshuf), _window_frac(glp)
+    assert shuf < 0.5 * nat          # shuffle really destroyed locality
+    assert rec > shuf + 0.5 * (nat - shuf)  # LP recovered >half the gap
+
+
+def test_apply_ordering_training_equivalent():
+    # full-graph training is permutation-equivariant: same loss trajectory
+    # (fp32, no dropout) on the relabeled dataset
+    from roc_amd.graph import apply_ordering, synthetic_dataset
+    from roc_amd import build_shard, build_model, AdamOptimizer, Trainer
+    from roc_amd.parallel.partition import edge_balanced_bounds
+
+    def run(g, feats, labels, mask, c):
+        torch.manual_seed(0)
+        shard = build_shard(g, 0, 1, edge_balanced_bounds(g.rowptr, 1))
+        model = build_model("gcn", [feats.shape[1], 16, c], dropout=0.0,
+                            seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.01)
+        tr = Trainer(model, shard, feats, labels, mask, opt)
+        return [tr.train_epoch() for _ in range(3)]
+
+    g, feats, labels, mask, c = synthetic_dataset("cora", seed=4)
+    base = run(g, feats, labels, mask, c)
+    for kind in ("degree", "rcm", "cluster"):
+        g2, f2, l2, m2, perm = apply_ordering(g, feats, labels, mask, kind)
+        assert sorted(perm.tolist()) == list(range(g.num_nodes))
+        out = run(g2, f2, l2, m2, c)
+        for a, b in zip(base, out):  # metrics[1] = CE loss sum
+            la, lb = float(a[1]), float(b[1])
+            assert abs(la - lb) < 1e-3 * max(1.0, abs(la))

@@ -68,6 +68,12 @@ def parse_args():
     ap.add_argument("--learnable-labels", action="store_true",
                     help="synthetic labels from a random one-hop teacher "
                          "(accuracy can actually rise)")
+    ap.add_argument("--reorder", default="none",
+                    choices=["none", "degree", "rcm", "cluster"],
+                    help="locality relabeling applied to the whole dataset "
+                         "before partitioning (sparser halos, denser SpMM "
+                         "gathers); not available with windowed --file "
+                         "loading (world>1 + --file)")
     ap.add_argument("--rebalance-every", type=int, default=0,
                     help="cost-model repartition every N epochs (measured "
                          "per-rank throughput; multi-rank only)")
@@ -142,6 +148,13 @@ def main():
 
     (g, feats, labels, mask, num_classes), pre_shard = \
         load_dataset(args, rank, world)
+    if args.reorder != "none":
+        if g is None:
+            sys.exit("--reorder needs the whole graph in memory; it is not "
+                     "available with windowed --file loading (world>1)")
+        from roc_amd.graph import apply_ordering
+        g, feats, labels, mask, _ = apply_ordering(
+            g, feats, labels, mask, args.reorder)
     # pad feature dim for 16-B-aligned bf16 rows (zero cols; exact math)
     pad = (-feats.shape[1]) % 8
     if pad and on_gpu:
