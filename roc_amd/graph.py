@@ -418,18 +418,23 @@ def _learnable_labels(g: CSRGraph, feats: torch.Tensor, c: int,
 
 
 def synthetic_dataset(name: str, seed: int = 1, scale: float = 1.0,
-                      learnable_labels: bool = False):
+                      learnable_labels: bool = False, locality: float = 0.0,
+                      num_communities: int = 64):
     """Graph + features + labels + masks of the named shape.
 
     ``scale`` < 1 shrinks nodes/edges proportionally (for tests).
     ``learnable_labels`` draws labels from a random one-hop teacher
     instead of uniformly (same shapes/cost; lets accuracy actually rise).
+    ``locality``/``num_communities``: planted community structure (see
+    ``synthetic_graph``) — pair with a random relabel + ``--reorder`` to
+    study ordering effects, or use directly for sparse-halo runs.
     Returns (graph, features fp32 [N, in_dim], labels int64 [N], mask int32 [N]).
     """
     n, e, d, c = DATASET_SHAPES[name]
     n = max(int(n * scale), 16)
     e = max(int(e * scale), n)
-    g = synthetic_graph(n, e, seed=seed)
+    g = synthetic_graph(n, e, seed=seed, locality=locality,
+                        num_communities=num_communities)
     rng = np.random.default_rng(seed + 1)
     feats = torch.from_numpy(rng.standard_normal((n, d), dtype=np.float32))
     if learnable_labels:

@@ -68,6 +68,12 @@ def parse_args():
     ap.add_argument("--learnable-labels", action="store_true",
                     help="synthetic labels from a random one-hop teacher "
                          "(accuracy can actually rise)")
+    ap.add_argument("--locality", type=float, default=0.0,
+                    help="synthetic graphs: fraction of edges kept inside "
+                         "a planted community (community structure like "
+                         "real graphs; see --reorder)")
+    ap.add_argument("--communities", type=int, default=64,
+                    help="synthetic graphs: number of planted communities")
     ap.add_argument("--reorder", default="none",
                     choices=["none", "degree", "rcm", "cluster"],
                     help="locality relabeling applied to the whole dataset "
@@ -128,7 +134,9 @@ def load_dataset(args, rank=0, world=1):
     name = args.dataset.replace("-synthetic", "")
     assert name in DATASET_SHAPES, f"unknown dataset {name}"
     return synthetic_dataset(name, seed=args.seed, scale=args.scale,
-                             learnable_labels=args.learnable_labels), None
+                             learnable_labels=args.learnable_labels,
+                             locality=args.locality,
+                             num_communities=args.communities), None
 
 
 def main():
