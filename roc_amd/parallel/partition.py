@@ -22,7 +22,26 @@ from typing import List, Optional
 import numpy as np
 import torch
 
+import os
+
 from ..graph import CSRGraph, build_transpose
+
+
+def _gather_locality(rowptr: np.ndarray, colidx: np.ndarray,
+                     max_sample: int = 1_000_000) -> float:
+    """Sampled fraction of edges whose source lies within a 16k-row
+    window of the destination (±8192 rows ≈ one XCD's 4 MiB L2 at
+    D=256 bf16). High → the graph is community-ordered and the natural
+    row schedule keeps consecutive workgroups in the same L2 window."""
+    ne = int(colidx.shape[0])
+    if ne == 0:
+        return 0.0
+    if ne > max_sample:
+        e = np.linspace(0, ne - 1, max_sample).astype(np.int64)
+    else:
+        e = np.arange(ne, dtype=np.int64)
+    rows = np.searchsorted(rowptr, e, side="right") - 1
+    return float((np.abs(rows - colidx[e].astype(np.int64)) < 8192).mean())
 
 
 @dataclass
@@ -45,8 +64,8 @@ class GraphShard:
     rsqrt_deg_local: torch.Tensor
     inv_deg_local: torch.Tensor
     rsqrt_deg_ext: torch.Tensor  # fp32 [n_ext] (local ++ halo)
-    row_order: torch.Tensor      # int32 [n_local] degree-descending
-    t_row_order: torch.Tensor    # int32 [n_ext]
+    row_order: Optional[torch.Tensor]    # int32 [n_local] degree-desc, or
+    t_row_order: Optional[torch.Tensor]  # None = natural schedule (r16)
     halo_ids: torch.Tensor       # int64 [n_halo] global ids (grouped by owner)
     recv_splits: List[int]       # rows received from each rank
     send_idx: torch.Tensor       # int64 local indices to send (concat by dest)
@@ -254,11 +273,26 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
     deg_local = deg_all[lo:hi]
     deg_ext = np.concatenate([deg_local, deg_all[halo_ids]]) if n_halo else deg_local
 
-    # degree-descending row orders (long rows scheduled first in the kernel)
+    # Row scheduling for the SpMM kernels. Two regimes (measured,
+    # profiles/r16): on a community-ordered graph the NATURAL order wins
+    # ~10% — consecutive workgroups gather from the same L2-scale source
+    # window — while on a locality-free ordering the global
+    # degree-descending order wins ~7% (hub rows scheduled first, no tail).
+    # Auto-pick by a sampled in-window gather fraction; override with
+    # ROC_SPMM_SCHEDULE=degree|natural.
     local_deg_edges = np.diff(local_rowptr)
-    row_order = np.argsort(-local_deg_edges, kind="stable").astype(np.int32)
-    t_deg = np.diff(t_rowptr.numpy())
-    t_row_order = np.argsort(-t_deg, kind="stable").astype(np.int32)
+    sched = os.environ.get("ROC_SPMM_SCHEDULE", "auto")
+    if sched == "auto":
+        sched = ("natural"
+                 if _gather_locality(local_rowptr, colidx) >= 0.5
+                 else "degree")
+    if sched == "natural":
+        row_order = None
+        t_row_order = None
+    else:
+        row_order = np.argsort(-local_deg_edges, kind="stable").astype(np.int32)
+        t_deg = np.diff(t_rowptr.numpy())
+        t_row_order = np.argsort(-t_deg, kind="stable").astype(np.int32)
 
     shard_kw = {}
     if world_size > 1:
@@ -337,8 +371,10 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
         rsqrt_deg_local=torch.from_numpy(1.0 / np.sqrt(deg_local)),
         inv_deg_local=torch.from_numpy(1.0 / deg_local),
         rsqrt_deg_ext=torch.from_numpy(1.0 / np.sqrt(deg_ext)),
-        row_order=torch.from_numpy(row_order),
-        t_row_order=torch.from_numpy(t_row_order),
+        row_order=(torch.from_numpy(row_order)
+                   if row_order is not None else None),
+        t_row_order=(torch.from_numpy(t_row_order)
+                     if t_row_order is not None else None),
         halo_ids=torch.from_numpy(halo_ids),
         recv_splits=recv_splits,
         send_idx=send_idx,

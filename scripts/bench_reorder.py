@@ -25,13 +25,24 @@ import numpy as np
 import torch
 
 
-def time_spmm(g, D, rounds, dev="cuda:0"):
+def time_spmm(g, D, rounds, schedule="degree", dev="cuda:0"):
+    """schedule: 'degree' (global degree-desc row order), 'natural'
+    (no reorder — preserves community grouping in launch order), or
+    'degree-block' (degree-desc WITHIN each 4096-row block — wide-row
+    balance AND block locality)."""
     from roc_amd import _C
     rowptr = g.rowptr.to(dev)
     colidx = g.colidx.to(dev)
     deg = (g.rowptr[1:] - g.rowptr[:-1]).float().clamp(min=1)
     rsq = deg.rsqrt().to(dev)
-    row_order = torch.argsort(-deg).int().to(dev)
+    if schedule == "degree":
+        row_order = torch.argsort(-deg).int().to(dev)
+    elif schedule == "degree-block":
+        blk = torch.arange(g.num_nodes, dtype=torch.float64) // 4096
+        key = blk * 1e9 - deg.double()  # block-major, degree-desc inside
+        row_order = torch.argsort(key).int().to(dev)
+    else:
+        row_order = None
     x = torch.randn(g.num_nodes, D).to(torch.bfloat16).to(dev)
     out = torch.empty_like(x)
     s = torch.cuda.Event(enable_timing=True)
@@ -65,6 +76,8 @@ def main():
     ap.add_argument("--dims", type=int, nargs="+", default=[256])
     ap.add_argument("--orderings", nargs="+",
                     default=["natural", "shuffled", "cluster", "rcm"])
+    ap.add_argument("--schedules", nargs="+",
+                    default=["degree", "natural", "degree-block"])
     args = ap.parse_args()
 
     from roc_amd.graph import (synthetic_graph, reorder_graph, ORDERINGS)
@@ -102,13 +115,14 @@ def main():
                                   "frac_in_64k_window": round(wf, 3)}),
                       flush=True)
                 continue
-            ms = time_spmm(g, D, args.rounds)
-            r = {"ordering": name, "D": D, "ms": round(ms, 2),
-                 "eff_TBs": round(gb / ms, 2),
-                 "frac_in_64k_window": round(wf, 3),
-                 "ordering_cost_s": round(order_s, 1)}
-            results.append(r)
-            print(json.dumps(r), flush=True)
+            for sched in args.schedules:
+                ms = time_spmm(g, D, args.rounds, schedule=sched)
+                r = {"ordering": name, "schedule": sched, "D": D,
+                     "ms": round(ms, 2), "eff_TBs": round(gb / ms, 2),
+                     "frac_in_64k_window": round(wf, 3),
+                     "ordering_cost_s": round(order_s, 1)}
+                results.append(r)
+                print(json.dumps(r), flush=True)
 
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/reorder_ab.json", "w") as f:

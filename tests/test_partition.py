@@ -95,3 +95,28 @@ def test_rebalance_shifts_toward_fast_rank():
     e0_old = rp[bounds[1]] - rp[bounds[0]]
     e0_new = rp[nb[1]] - rp[nb[0]]
     assert e0_new < e0_old
+
+
+def test_spmm_schedule_auto():
+    # community-ordered graph -> natural schedule (row_order None);
+    # scrambled ids -> degree-descending order; env forces either way
+    import os
+    import numpy as np
+    from roc_amd.graph import synthetic_graph, reorder_graph
+    from roc_amd.parallel.partition import build_shard
+    g = synthetic_graph(60000, 600_000, seed=11, locality=0.9,
+                        num_communities=15)
+    rng = np.random.default_rng(2)
+    gshuf = reorder_graph(g, torch.from_numpy(rng.permutation(60000)))
+    assert build_shard(g, 0, 1).row_order is None
+    sh = build_shard(gshuf, 0, 1)
+    assert sh.row_order is not None
+    deg = (sh.rowptr[1:] - sh.rowptr[:-1])[sh.row_order.long()]
+    assert (deg[:-1] >= deg[1:]).all()  # degree-descending
+    try:
+        os.environ["ROC_SPMM_SCHEDULE"] = "degree"
+        assert build_shard(g, 0, 1).row_order is not None
+        os.environ["ROC_SPMM_SCHEDULE"] = "natural"
+        assert build_shard(gshuf, 0, 1).row_order is None
+    finally:
+        del os.environ["ROC_SPMM_SCHEDULE"]
