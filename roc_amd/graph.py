@@ -35,6 +35,7 @@ MASK_VAL = 2
 MASK_TEST = 3
 
 _MASK_STR = {"None": MASK_NONE, "Train": MASK_TRAIN, "Val": MASK_VAL, "Test": MASK_TEST}
+MASK_NAMES = {v: k for k, v in _MASK_STR.items()}
 
 
 @dataclass

@@ -174,3 +174,44 @@ def test_all_config_yamls_run(tmp_path):
             cwd=repo, env=env, capture_output=True, text=True, timeout=300)
         assert r.returncode == 0, (cfg, r.stderr[-1500:])
         assert "[config]" in r.stdout, (cfg, r.stdout[-500:])
+
+
+def test_reorder_dataset_tool(tmp_path):
+    # offline preprocessing: scripts/reorder_dataset.py writes a relabeled
+    # reference-format dataset whose aggregation commutes with the perm
+    import subprocess
+    import sys
+    import numpy as np
+    from roc_amd.graph import (synthetic_dataset, save_lux, MASK_NAMES,
+                               load_lux, load_features, load_labels,
+                               load_mask)
+    from roc_amd.ops import functional as F
+    from roc_amd.parallel.partition import build_shard
+    g, feats, labels, mask, c = synthetic_dataset("cora", seed=2, scale=0.2)
+    pre = str(tmp_path / "cora")
+    save_lux(pre + ".add_self_edge.lux", g)
+    feats.numpy().astype(np.float32).tofile(pre + ".feats.bin")
+    with open(pre + ".label", "w") as f:
+        f.write("\n".join(str(int(v)) for v in labels) + "\n")
+    with open(pre + ".mask", "w") as f:
+        f.write("\n".join(MASK_NAMES[int(v)] for v in mask) + "\n")
+    import roc_amd
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(roc_amd.__file__)))
+    out = str(tmp_path / "cora-lp")
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts/reorder_dataset.py"),
+         "--in", pre, "--out", out, "--order", "cluster",
+         "--in-dim", str(feats.shape[1])],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    g2 = load_lux(out + ".add_self_edge.lux")
+    f2 = load_features(out, g2.num_nodes, feats.shape[1])
+    perm = torch.from_numpy(np.load(out + ".perm.npy"))
+    assert torch.equal(f2, feats[perm])
+    assert torch.equal(load_labels(out + ".label", g2.num_nodes),
+                       labels[perm])
+    assert torch.equal(load_mask(out + ".mask", g2.num_nodes), mask[perm])
+    x = torch.randn(g.num_nodes, 5)
+    o1 = F.scatter_gather(x, build_shard(g, 0, 1))
+    o2 = F.scatter_gather(x[perm], build_shard(g2, 0, 1))
+    assert torch.allclose(o2, o1[perm], atol=1e-5)
