@@ -41,6 +41,16 @@ def time_spmm(g, D, rounds, schedule="degree", dev="cuda:0"):
         blk = torch.arange(g.num_nodes, dtype=torch.float64) // 4096
         key = blk * 1e9 - deg.double()  # block-major, degree-desc inside
         row_order = torch.argsort(key).int().to(dev)
+    elif schedule == "xcd-split":
+        # round-robin dispatch sends consecutive workgroups to different
+        # XCDs; interleave 8 contiguous regions so XCD k streams region k
+        # (each XCD L2 holds ITS OWN window instead of 8 copies of one)
+        n = g.num_nodes
+        chunk = (n + 7) // 8
+        pad = np.full(chunk * 8, -1, dtype=np.int64)
+        pad[:n] = np.arange(n)
+        ro = pad.reshape(8, chunk).T.reshape(-1)
+        row_order = torch.from_numpy(ro[ro >= 0].astype(np.int32)).to(dev)
     else:
         row_order = None
     x = torch.randn(g.num_nodes, D).to(torch.bfloat16).to(dev)
