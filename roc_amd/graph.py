@@ -297,11 +297,26 @@ ORDERINGS = {"degree": degree_order, "rcm": rcm_order,
 
 def apply_ordering(g: CSRGraph, feats: torch.Tensor, labels: torch.Tensor,
                    mask: torch.Tensor, kind: str):
-    """Relabel the whole dataset by a named locality ordering ("degree" or
-    "rcm"); returns (graph, feats, labels, mask, perm). Deterministic, so
-    every rank computes the identical permutation. Training on the
-    relabeled dataset is mathematically the same full-graph problem
-    (permutation equivariance)."""
+    """Relabel the whole dataset by a named locality ordering ("degree",
+    "rcm", "cluster", or "auto"); returns (graph, feats, labels, mask,
+    perm — None if "auto" kept the original). Deterministic, so every
+    rank computes the identical permutation. Training on the relabeled
+    dataset is mathematically the same full-graph problem (permutation
+    equivariance).
+
+    "auto": try the LP-cluster ordering and keep it only if it
+    meaningfully improves the sampled gather locality (the signal the
+    SpMM schedule and the halo plan actually respond to — profiles/r16);
+    a genuinely locality-free graph is returned unchanged."""
+    if kind == "auto":
+        from .parallel.partition import _gather_locality
+        before = _gather_locality(g.rowptr.numpy(), g.colidx.numpy())
+        perm = cluster_order(g)
+        g2 = reorder_graph(g, perm)
+        after = _gather_locality(g2.rowptr.numpy(), g2.colidx.numpy())
+        if after < 0.5 or after < 2 * before:
+            return g, feats, labels, mask, None
+        return g2, feats[perm], labels[perm], mask[perm], perm
     perm = ORDERINGS[kind](g)
     return (reorder_graph(g, perm), feats[perm], labels[perm], mask[perm],
             perm)

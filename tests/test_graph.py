@@ -219,3 +219,25 @@ def test_cluster_order_shrinks_halo():
 
     h_shuf, h_lp = halo_rows(gshuf), halo_rows(glp)
     assert h_lp < 0.6 * h_shuf, (h_shuf, h_lp)
+
+
+def test_apply_ordering_auto():
+    # auto: reorders a scrambled community graph, leaves a uniform graph
+    # (and an already-ordered one) alone
+    # (graph must be ≫ the ±8192-row locality window to discriminate)
+    from roc_amd.graph import apply_ordering, reorder_graph
+    rng = np.random.default_rng(3)
+    n = 100_000
+    gc = synthetic_graph(n, 1_000_000, seed=3, locality=0.85,
+                         num_communities=25)
+    feats = torch.randn(n, 4)
+    labels = torch.zeros(n, dtype=torch.int64)
+    mask = torch.ones(n, dtype=torch.int32)
+    gshuf = reorder_graph(gc, torch.from_numpy(rng.permutation(n)))
+    _, _, _, _, perm = apply_ordering(gshuf, feats, labels, mask, "auto")
+    assert perm is not None  # scrambled community graph -> reordered
+    _, _, _, _, perm = apply_ordering(gc, feats, labels, mask, "auto")
+    assert perm is None      # already ordered -> kept
+    gu = synthetic_graph(n, 1_000_000, seed=4)
+    _, _, _, _, perm = apply_ordering(gu, feats, labels, mask, "auto")
+    assert perm is None      # uniform -> kept

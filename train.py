@@ -75,7 +75,7 @@ def parse_args():
     ap.add_argument("--communities", type=int, default=64,
                     help="synthetic graphs: number of planted communities")
     ap.add_argument("--reorder", default="none",
-                    choices=["none", "degree", "rcm", "cluster"],
+                    choices=["none", "degree", "rcm", "cluster", "auto"],
                     help="locality relabeling applied to the whole dataset "
                          "before partitioning (sparser halos, denser SpMM "
                          "gathers); not available with windowed --file "
