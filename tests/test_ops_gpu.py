@@ -398,3 +398,24 @@ def test_ag_split_spmm_blocks_match_reference():
     want_full = ref.spmm(dy, sh.ag_t_rowptr, sh.ag_t_colidx, ws * mr)
     assert torch.allclose(dfull.cpu(), want_full, rtol=1e-4, atol=1e-3), \
         (dfull.cpu() - want_full).abs().max()
+
+
+def test_spmm_schedule_equivalence():
+    # the row schedule (natural vs degree-desc vs any permutation) is a
+    # pure performance knob: each row is reduced by one team in the same
+    # in-row order, so outputs must be BIT-identical
+    from roc_amd import _C
+    g = synthetic_graph(40000, 800_000, seed=13, locality=0.7,
+                        num_communities=10)
+    D = 128
+    rowptr = g.rowptr.to(DEV)
+    colidx = g.colidx.to(DEV)
+    deg = (g.rowptr[1:] - g.rowptr[:-1]).float().clamp(min=1)
+    rsq = deg.rsqrt().to(DEV)
+    x = torch.randn(g.num_nodes, D).to(torch.bfloat16).to(DEV)
+    out_nat = torch.empty_like(x)
+    out_deg = torch.empty_like(x)
+    _C.spmm(out_nat, x, rowptr, colidx, rsq, None, None)
+    order = torch.argsort(-deg).int().to(DEV)
+    _C.spmm(out_deg, x, rowptr, colidx, rsq, None, order)
+    assert torch.equal(out_nat, out_deg)
