@@ -181,7 +181,9 @@ torch::Tensor lp_cluster_order(torch::Tensor rowptr, torch::Tensor colidx,
       }
     }
     lab.swap(next);
-    if (changed.load(std::memory_order_relaxed) == 0) break;
+    // converged-enough: a vanishing tail of flips no longer moves the
+    // ordering (matters at 10^9-edge scale where each round is minutes)
+    if (changed.load(std::memory_order_relaxed) <= n / 4096) break;
   }
 
   auto out = torch::empty({n}, torch::kInt64);
