@@ -121,3 +121,30 @@ def test_sharded_training_matches_single_rank():
     md = res[0][1]
     assert md["train_total"] == m_single["train_total"]
     assert abs(md["ce_loss"] - m_single["ce_loss"]) < 1e-3
+
+
+def test_bench_contract_ws2_gloo():
+    """The driver's scale run launches bench.py under torchrun with
+    --gpus N: guard that contract end-to-end on CPU (gloo, ws=2) — one
+    JSON line on rank 0 with the whole-job metric fields."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import roc_amd
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(roc_amd.__file__)))
+    env = dict(os.environ)
+    env.pop("ROC_SPMM_SCHEDULE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29737", os.path.join(repo, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1", "--scale", "0.02"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 2 and out["steps"] == 2
+    assert out["unit"] == "s/epoch" and out["value"] > 0
+    assert out["higher_is_better"] is False
+    assert "parallelism" in out["config"]
