@@ -145,10 +145,12 @@ class Trainer:
     def train_epoch(self):
         self.model.train()
         if (self.use_graph and self.device.type == "cuda"
-                and self.tracer is None and self.offload is None):
+                and self.tracer is None and self.offload is None
+                and not getattr(self.model, "recompute", False)):
             metrics = self._graph_epoch()
             self.epoch += 1
             return metrics
+        F.next_dropout_epoch()
         if self.tracer is None:
             self.optimizer.zero_grad()
             loss, metrics = self._forward_loss()
@@ -200,6 +202,9 @@ class Trainer:
         self.model.eval()
 
     def zero_gradients(self):
+        # once-per-epoch in the reference loop (`gnn.cc:103`) — the right
+        # place to advance the dropout epoch base, matching train_epoch
+        F.next_dropout_epoch()
         self.optimizer.zero_grad()
 
     def forward(self):

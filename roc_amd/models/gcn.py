@@ -36,32 +36,46 @@ class GCN(torch.nn.Module):
             else:
                 self.res_proj.append(torch.nn.Parameter(torch.empty(0)))
 
+    # Activation recompute (capacity tier): when True, each layer's
+    # forward is re-run during backward instead of keeping its
+    # intermediates resident — O(1 layer) transient HBM instead of
+    # O(depth). Dropout stays exact via the per-layer Philox call_id.
+    recompute = False
+
+    def _layer(self, i, x, shard, group):
+        w = self.weights[i]
+        h = F.dropout(x, self.p, self.training, call_id=i)
+        if self.fused_norm:
+            # source-side D^-1/2 rides the GEMM epilogue (owner rank
+            # scales its rows BEFORE the halo exchange); dst-side
+            # D^-1/2 rides the SpMM store. No per-edge degree gather.
+            h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
+            # strategy (halo a2a / overlap / all_gather) is the
+            # shard's choice — see parallel/aggregate.py
+            h = aggregate(h, shard, dst_scale=shard.rsqrt_deg_local,
+                          group=group)
+        else:
+            h = F.linear(h, w)
+            h = halo_exchange(h, shard, group)
+            h = F.indegree_norm(
+                F.scatter_gather(F.degree_scale(h, shard.rsqrt_deg_ext), shard),
+                shard)
+        if i < len(self.weights) - 1:
+            h = F.relu(h)
+        if self.residual:
+            proj = self.res_proj[i]
+            if proj.numel() > 0:
+                h = F.add(h, F.linear(x, proj))
+            elif x.shape == h.shape:
+                h = F.add(h, x)
+        return h
+
     def forward(self, x, shard, group=None):
-        nlayers = len(self.weights)
-        for i, w in enumerate(self.weights):
-            h = F.dropout(x, self.p, self.training)
-            if self.fused_norm:
-                # source-side D^-1/2 rides the GEMM epilogue (owner rank
-                # scales its rows BEFORE the halo exchange); dst-side
-                # D^-1/2 rides the SpMM store. No per-edge degree gather.
-                h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
-                # strategy (halo a2a / overlap / all_gather) is the
-                # shard's choice — see parallel/aggregate.py
-                h = aggregate(h, shard, dst_scale=shard.rsqrt_deg_local,
-                              group=group)
+        for i in range(len(self.weights)):
+            if self.recompute and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    self._layer, i, x, shard, group,
+                    use_reentrant=False, preserve_rng_state=False)
             else:
-                h = F.linear(h, w)
-                h = halo_exchange(h, shard, group)
-                h = F.indegree_norm(
-                    F.scatter_gather(F.degree_scale(h, shard.rsqrt_deg_ext), shard),
-                    shard)
-            if i < nlayers - 1:
-                h = F.relu(h)
-            if self.residual:
-                proj = self.res_proj[i]
-                if proj.numel() > 0:
-                    h = F.add(h, F.linear(x, proj))
-                elif x.shape == h.shape:
-                    h = F.add(h, x)
-            x = h
+                x = self._layer(i, x, shard, group)
         return x

@@ -29,15 +29,24 @@ class GIN(torch.nn.Module):
             self.w2.append(torch.nn.Parameter(
                 glorot_uniform((hid, dims[i + 1]), seed=seed + 2 * i + 1)))
 
+    recompute = False  # see GCN.recompute
+
+    def _layer(self, i, x, shard, group):
+        h = F.dropout(x, self.p, self.training, call_id=i)
+        agg = aggregate(h, shard, group=group)
+        h = agg + (1.0 + self.eps[i]).to(h.dtype) * h  # scalar-eps glue
+        h = F.linear(h, self.w1[i], activation="relu")
+        h = F.linear(h, self.w2[i])
+        if i < len(self.w1) - 1:
+            h = F.relu(h)
+        return h
+
     def forward(self, x, shard, group=None):
-        nlayers = len(self.w1)
-        for i in range(nlayers):
-            h = F.dropout(x, self.p, self.training)
-            agg = aggregate(h, shard, group=group)
-            h = agg + (1.0 + self.eps[i]).to(h.dtype) * h  # scalar-eps glue
-            h = F.linear(h, self.w1[i], activation="relu")
-            h = F.linear(h, self.w2[i])
-            if i < nlayers - 1:
-                h = F.relu(h)
-            x = h
+        for i in range(len(self.w1)):
+            if self.recompute and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    self._layer, i, x, shard, group,
+                    use_reentrant=False, preserve_rng_state=False)
+            else:
+                x = self._layer(i, x, shard, group)
         return x

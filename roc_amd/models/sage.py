@@ -30,15 +30,24 @@ class GraphSAGE(torch.nn.Module):
                 glorot_uniform((dims[i], dims[i + 1]), seed=seed + 2 * i + 1)))
 
     def forward(self, x, shard, group=None):
-        nlayers = len(self.w_self)
-        for i in range(nlayers):
-            h = F.dropout(x, self.p, self.training)
-            h_self = F.linear(h, self.w_self[i])
-            hn = F.linear(h, self.w_neigh[i])
-            hn = aggregate(hn, shard, dst_scale=shard.inv_deg_local,
-                           group=group)  # fused mean
-            h = F.add(h_self, hn)
-            if i < nlayers - 1:
-                h = F.relu(h)
-            x = h
+        for i in range(len(self.w_self)):
+            if self.recompute and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    self._layer, i, x, shard, group,
+                    use_reentrant=False, preserve_rng_state=False)
+            else:
+                x = self._layer(i, x, shard, group)
         return x
+
+    recompute = False  # see GCN.recompute
+
+    def _layer(self, i, x, shard, group):
+        h = F.dropout(x, self.p, self.training, call_id=i)
+        h_self = F.linear(h, self.w_self[i])
+        hn = F.linear(h, self.w_neigh[i])
+        hn = aggregate(hn, shard, dst_scale=shard.inv_deg_local,
+                       group=group)  # fused mean
+        h = F.add(h_self, hn)
+        if i < len(self.w_self) - 1:
+            h = F.relu(h)
+        return h

@@ -317,7 +317,7 @@ def mul(a, b):
 # tensor is ever materialized on GPU)
 # ---------------------------------------------------------------------------
 
-_DROPOUT_STATE = {"seed": 1, "offset": 0, "counter": None}
+_DROPOUT_STATE = {"seed": 1, "offset": 0, "counter": None, "epoch": 0}
 _WEIGHT_VERSION = [0]
 
 
@@ -346,6 +346,7 @@ def _cast_weight(w, dtype):
 def set_dropout_seed(seed: int) -> None:
     _DROPOUT_STATE["seed"] = int(seed)
     _DROPOUT_STATE["offset"] = 0
+    _DROPOUT_STATE["epoch"] = 0
 
 
 def set_dropout_counter(counter) -> None:
@@ -359,13 +360,30 @@ def reset_dropout_offset() -> None:
     _DROPOUT_STATE["offset"] = 0
 
 
+def next_dropout_epoch() -> None:
+    """Advance the per-epoch Philox base for call_id-addressed dropout
+    (eager path; the hipGraph path gets its epoch from the device
+    counter instead). Masks then vary per epoch while staying exactly
+    reproducible WITHIN the epoch — the property activation recompute
+    needs."""
+    _DROPOUT_STATE["epoch"] += 1
+
+
 class _Dropout(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, p):
+    def forward(ctx, x, p, call_id=None):
         seed = _DROPOUT_STATE["seed"]
-        offset = _DROPOUT_STATE["offset"]
         counter = _DROPOUT_STATE["counter"]
-        _DROPOUT_STATE["offset"] += 1
+        if call_id is not None:
+            # stable per-epoch offset: a checkpointed layer re-running its
+            # forward during backward regenerates the IDENTICAL mask.
+            # Epoch variation: device counter (graph mode) or host epoch.
+            offset = int(call_id)
+            if counter is None:
+                offset += _DROPOUT_STATE["epoch"] * 65536
+        else:
+            offset = _DROPOUT_STATE["offset"]
+            _DROPOUT_STATE["offset"] += 1
         ctx.params = (p, seed, offset, counter)
         if _hip(x):
             y = torch.empty_like(x)
@@ -384,14 +402,19 @@ class _Dropout(torch.autograd.Function):
             _C.dropout_fwd(dx, dy, p, seed, offset, counter)  # same stream
             return dx, None
         (mask,) = ctx.saved_tensors
-        return dy * mask.to(dy.dtype) / (1.0 - p), None
+        return dy * mask.to(dy.dtype) / (1.0 - p), None, None
 
 
-def dropout(x, p: float, training: bool):
-    """Train: mask+scale; infer: identity (reference `dropout_kernel.cu:159-180`)."""
+def dropout(x, p: float, training: bool, call_id=None):
+    """Train: mask+scale; infer: identity (reference `dropout_kernel.cu:159-180`).
+
+    ``call_id``: stable per-epoch Philox offset (= the model's layer
+    index). Required under activation recompute; numerically identical
+    to the global running offset for the shipped models (one dropout
+    per layer, called in layer order)."""
     if not training or p <= 0.0:
         return x
-    return _Dropout.apply(x, p)
+    return _Dropout.apply(x, p, call_id)
 
 
 # ---------------------------------------------------------------------------

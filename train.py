@@ -65,6 +65,10 @@ def parse_args():
                     help="write a chrome trace JSON here at the end")
     ap.add_argument("--offload", action="store_true",
                     help="host-DRAM activation offload (capacity tier)")
+    ap.add_argument("--recompute", action="store_true",
+                    help="per-layer activation recompute (capacity tier: "
+                         "O(1-layer) transient HBM, no pinned host memory; "
+                         "exact — dropout masks regenerate identically)")
     ap.add_argument("--learnable-labels", action="store_true",
                     help="synthetic labels from a random one-hop teacher "
                          "(accuracy can actually rise)")
@@ -192,6 +196,10 @@ def main():
     mkw = {"residual": args.residual} if args.model == "gcn" else {}
     model = build_model(args.model, dims, dropout=args.dropout,
                         seed=args.seed, **mkw)
+    if args.recompute:
+        assert not args.offload, "--recompute and --offload are exclusive " \
+            "(both re-route activation residency)"
+        model.recompute = True
     opt = AdamOptimizer(model.parameters(), lr=args.lr,
                         weight_decay=args.weight_decay,
                         decay_rate=args.decay_rate,
@@ -261,6 +269,11 @@ def main():
                 and ep % args.checkpoint_every == 0):
             save_checkpoint(args.checkpoint, trainer)
 
+    if rank == 0 and on_gpu:
+        print(f"[mem] peak HBM allocated "
+              f"{torch.cuda.max_memory_allocated() / 2**30:.1f} GiB / "
+              f"reserved {torch.cuda.max_memory_reserved() / 2**30:.1f} GiB",
+              flush=True)
     if args.checkpoint:
         save_checkpoint(args.checkpoint, trainer)
     if args.trace and trainer.tracer is not None:
