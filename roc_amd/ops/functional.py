@@ -400,7 +400,7 @@ class _Dropout(torch.autograd.Function):
         if _hip(dy):
             dx = torch.empty_like(dy)
             _C.dropout_fwd(dx, dy, p, seed, offset, counter)  # same stream
-            return dx, None
+            return dx, None, None
         (mask,) = ctx.saved_tensors
         return dy * mask.to(dy.dtype) / (1.0 - p), None, None
 
