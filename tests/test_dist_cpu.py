@@ -148,3 +148,47 @@ def test_bench_contract_ws2_gloo():
     assert out["unit"] == "s/epoch" and out["value"] > 0
     assert out["higher_is_better"] is False
     assert "parallelism" in out["config"]
+
+
+def _recompute_worker(rank, port, q):
+    # sharded training with per-layer recompute: the collectives re-run
+    # INSIDE backward must line up identically across ranks
+    try:
+        _init(rank, port)
+
+        def run(recompute, comm_mode):
+            os.environ["ROC_COMM_MODE"] = comm_mode
+            torch.manual_seed(0)
+            g, feats, labels, mask, c = synthetic_dataset(
+                "cora", scale=0.05, seed=3)
+            bounds = edge_balanced_bounds(g.rowptr, WS)
+            sh = build_shard(g, rank, WS, bounds)
+            model = build_model("gcn", [feats.shape[1], 16, 16, c],
+                                dropout=0.4, seed=1)
+            model.recompute = recompute
+            opt = AdamOptimizer(model.parameters(), lr=0.01)
+            tr = Trainer(model, sh, feats, labels, mask, opt)
+            for _ in range(2):
+                tr.train_epoch()
+            return model.weights[0].detach().numpy().copy()
+
+        for mode in ("halo", "allgather"):
+            w_std = run(False, mode)
+            w_rec = run(True, mode)
+            if not np.allclose(w_std, w_rec, atol=1e-6):
+                q.put((rank, None, None,
+                       f"{mode}: recompute diverged from standard"))
+                return
+        q.put((rank, True, None, None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, None, None, repr(e)))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_recompute_distributed_matches_standard():
+    res = _run(_recompute_worker, 29517)
+    for rank, ok, _, err in res:
+        assert err is None, f"rank {rank}: {err}"
+        assert ok
