@@ -16,13 +16,12 @@ before each aggregation. The plan below precomputes, once per graph:
 """
 from __future__ import annotations
 
+import os
 from dataclasses import dataclass
 from typing import List, Optional
 
 import numpy as np
 import torch
-
-import os
 
 from ..graph import CSRGraph, build_transpose
 
@@ -237,9 +236,8 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
     comm_mode = "none"
     halo_fraction = 0.0
     if world_size > 1:
-        import os as _os
         halo_fraction = n_halo / max(bounds[-1] - n_local, 1)
-        comm_mode = _os.environ.get("ROC_COMM_MODE", "auto")
+        comm_mode = os.environ.get("ROC_COMM_MODE", "auto")
         if comm_mode == "auto":
             comm_mode = "allgather" if halo_fraction > 0.5 else "halo"
 
