@@ -1,0 +1,71 @@
+"""Property-based invariants for the graph machinery (hypothesis):
+random CSR graphs -> transpose/reorder/ordering utilities must hold
+their structural contracts for ANY input, not just the fixture shapes."""
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from roc_amd.graph import (CSRGraph, build_transpose, reorder_graph,
+                           rcm_order, cluster_order, degree_order,
+                           synthetic_graph)
+
+
+@st.composite
+def graphs(draw):
+    n = draw(st.integers(min_value=1, max_value=120))
+    e = draw(st.integers(min_value=0, max_value=6 * n))
+    seed = draw(st.integers(min_value=0, max_value=2**31 - 1))
+    self_edges = draw(st.booleans())
+    loc = draw(st.sampled_from([0.0, 0.5, 0.9]))
+    return synthetic_graph(n, max(e, n if self_edges else 0), seed=seed,
+                           add_self_edges=self_edges, locality=loc,
+                           num_communities=draw(st.integers(1, 8)))
+
+
+def _degree_multiset(g):
+    return sorted(np.diff(g.rowptr.numpy()).tolist())
+
+
+@settings(max_examples=40, deadline=None)
+@given(graphs())
+def test_transpose_involution(g):
+    # (A^T)^T == A as a multigraph: edge multiset preserved
+    trp, tci = build_transpose(g.num_nodes, g.rowptr, g.colidx)
+    trp2, tci2 = build_transpose(g.num_nodes, trp, tci)
+    rows = np.repeat(np.arange(g.num_nodes), np.diff(g.rowptr.numpy()))
+    e1 = sorted(zip(rows.tolist(), g.colidx.numpy().tolist()))
+    rows2 = np.repeat(np.arange(g.num_nodes), np.diff(trp2.numpy()))
+    e2 = sorted(zip(rows2.tolist(), tci2.numpy().tolist()))
+    assert e1 == e2
+    assert int(trp[-1]) == g.num_edges
+
+
+@settings(max_examples=40, deadline=None)
+@given(graphs())
+def test_orderings_are_permutations_preserving_structure(g):
+    for fn in (degree_order, rcm_order, cluster_order):
+        perm = fn(g)
+        assert sorted(perm.tolist()) == list(range(g.num_nodes)), fn.__name__
+        g2 = reorder_graph(g, perm)
+        assert g2.num_edges == g.num_edges
+        # degree multiset invariant under relabeling
+        assert _degree_multiset(g2) == _degree_multiset(g), fn.__name__
+        # spot-check edge preservation: edge (u, v) <-> (pos(u), pos(v))
+        inv = np.empty(g.num_nodes, dtype=np.int64)
+        inv[perm.numpy()] = np.arange(g.num_nodes)
+        rows = np.repeat(np.arange(g.num_nodes), np.diff(g.rowptr.numpy()))
+        e1 = sorted(zip(inv[rows].tolist(),
+                        inv[g.colidx.numpy()].tolist()))
+        rows2 = np.repeat(np.arange(g.num_nodes), np.diff(g2.rowptr.numpy()))
+        e2 = sorted(zip(rows2.tolist(), g2.colidx.numpy().tolist()))
+        assert e1 == e2, fn.__name__
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(2, 60), st.integers(0, 2**31 - 1), st.integers(1, 6))
+def test_edge_balanced_bounds_cover(n, seed, parts):
+    from roc_amd.parallel.partition import edge_balanced_bounds
+    g = synthetic_graph(n, 4 * n, seed=seed)
+    b = edge_balanced_bounds(g.rowptr, parts)
+    assert b[0] == 0 and b[-1] == g.num_nodes
+    assert all(b[i] <= b[i + 1] for i in range(len(b) - 1))
