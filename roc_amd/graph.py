@@ -226,6 +226,15 @@ def reorder_graph(g: CSRGraph, perm: torch.Tensor) -> CSRGraph:
     ordering (degree sort, clustering) before partitioning real graphs —
     contiguous vertex ranges then cut fewer edges. Apply the same perm to
     features/labels/masks: x_new = x_old[perm]."""
+    try:
+        from roc_amd import _C
+        new_rp, new_ci = _C.csr_permute(g.rowptr.contiguous(),
+                                        g.colidx.contiguous(),
+                                        perm.to(torch.int64).contiguous())
+        return CSRGraph(num_nodes=g.num_nodes, num_edges=g.num_edges,
+                        rowptr=new_rp, colidx=new_ci)
+    except ImportError:
+        pass
     perm_np = perm.numpy()
     inv = np.empty_like(perm_np)
     inv[perm_np] = np.arange(g.num_nodes, dtype=perm_np.dtype)
