@@ -215,3 +215,17 @@ def test_reorder_dataset_tool(tmp_path):
     o1 = F.scatter_gather(x, build_shard(g, 0, 1))
     o2 = F.scatter_gather(x[perm], build_shard(g2, 0, 1))
     assert torch.allclose(o2, o1[perm], atol=1e-5)
+
+
+def test_bench_reorder_script_cpu_dryrun():
+    """scripts/bench_reorder.py degrades to a locality-fraction dry-run
+    without a GPU — guard the harness against bit-rot."""
+    import roc_amd
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(roc_amd.__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts/bench_reorder.py"),
+         "--nodes", "5000", "--edges", "50000", "--communities", "4",
+         "--orderings", "natural", "shuffled", "cluster"],
+        capture_output=True, text=True, timeout=300, cwd=repo)
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert r.stdout.count("frac_in_64k_window") >= 3
