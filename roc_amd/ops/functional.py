@@ -447,7 +447,10 @@ class _SoftmaxCE(torch.autograd.Function):
                  md["val_correct"], md["val_total"],
                  md["test_correct"], md["test_total"]],
                 dtype=torch.float32)
-            loss = metrics[1].clone()
+            # differentiable loss keeps the LOGITS dtype (an fp32 round-trip
+            # through the metrics container breaks fp64 gradcheck)
+            loss = torch.tensor(md["ce_loss"] * max(md["train_total"], 1),
+                                dtype=logits.dtype)
         ctx.save_for_backward(dl)
         ctx.mark_non_differentiable(metrics)
         return loss, metrics

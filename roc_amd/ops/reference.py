@@ -90,7 +90,11 @@ def softmax_cross_entropy(
             logits[:, :C].contiguous(), labels, mask, grad_scale)
         dl_full[:, :C] = dl
         return dl_full, md
-    lf = logits.to(torch.float32)
+    # fp32 accumulation for low-precision inputs (the HIP kernel's
+    # behavior); fp32/fp64 inputs keep their own precision (fp64 matters
+    # for gradcheck)
+    lf = logits if logits.dtype in (torch.float32, torch.float64) \
+        else logits.to(torch.float32)
     p = torch.softmax(lf, dim=1)
     n, c = lf.shape
     onehot = torch.zeros_like(p)

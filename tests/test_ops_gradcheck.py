@@ -3,7 +3,6 @@
 finite differences — including empty rows, hubs and multi-edges the
 fixed-shape numerics tests never draw."""
 import numpy as np
-import pytest
 import torch
 from hypothesis import given, settings, strategies as st
 
@@ -84,3 +83,33 @@ def test_elementwise_gradcheck(rows, d, seed):
     c.requires_grad_(True)
     assert torch.autograd.gradcheck(F.relu, (c,))
     assert torch.autograd.gradcheck(F.sigmoid, (a,))
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(2, 30), st.integers(2, 7), st.integers(0, 2**31 - 1))
+def test_softmax_ce_gradcheck(rows, classes, seed):
+    # mask-zeroed CE grad (softmax - onehot on Train rows only) vs finite
+    # differences of the returned loss
+    torch.manual_seed(seed)
+    rng = np.random.default_rng(seed)
+    labels = torch.from_numpy(rng.integers(0, classes, rows))
+    mask = torch.from_numpy(
+        rng.choice([0, 1, 2, 3], size=rows).astype(np.int32))
+    if (mask == 1).sum() == 0:
+        mask[0] = 1  # at least one Train row so the loss is non-trivial
+    logits = torch.randn(rows, classes, dtype=torch.float64,
+                         requires_grad=True)
+    assert torch.autograd.gradcheck(
+        lambda t: F.softmax_cross_entropy(t, labels, mask, 1.0)[0],
+        (logits,), eps=1e-6, atol=1e-5)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(1, 30), st.integers(1, 6), st.integers(0, 2**31 - 1))
+def test_dropout_gradcheck(rows, d, seed):
+    # fixed call_id -> fixed Philox mask -> differentiable given the mask
+    F.set_dropout_seed(seed)
+    x = torch.randn(rows, d, dtype=torch.float64, requires_grad=True)
+    assert torch.autograd.gradcheck(
+        lambda t: F.dropout(t, 0.4, True, call_id=3), (x,),
+        eps=1e-6, atol=1e-5)
