@@ -69,3 +69,54 @@ def test_edge_balanced_bounds_cover(n, seed, parts):
     b = edge_balanced_bounds(g.rowptr, parts)
     assert b[0] == 0 and b[-1] == g.num_nodes
     assert all(b[i] <= b[i + 1] for i in range(len(b) - 1))
+
+
+@settings(max_examples=25, deadline=None)
+@given(graphs(), st.integers(1, 5))
+def test_halo_plan_invariants(g, world):
+    """Structural contracts of the per-rank halo plan for ANY graph and
+    world size (no process group: the full-graph scan path)."""
+    import os
+    from roc_amd.parallel.partition import build_shard, edge_balanced_bounds
+    world = min(world, g.num_nodes)
+    bounds = edge_balanced_bounds(g.rowptr, world)
+    os.environ["ROC_COMM_MODE"] = "halo"  # allgather has no send plan
+    try:
+        shards = [build_shard(g, r, world, bounds) for r in range(world)]
+    finally:
+        del os.environ["ROC_COMM_MODE"]
+    assert bounds[0] == 0 and bounds[-1] == g.num_nodes
+
+    for sh in shards:
+        # halo ids are remote, unique, grouped by owner rank
+        halo = sh.halo_ids.numpy()
+        assert len(np.unique(halo)) == len(halo)
+        assert not ((halo >= sh.lo) & (halo < sh.hi)).any()
+        owners = np.searchsorted(np.asarray(bounds), halo, side="right") - 1
+        assert (np.diff(owners) >= 0).all()  # owner-major grouping
+        assert sum(sh.recv_splits) == sh.n_halo
+        # local CSR columns live in the ext space; transpose in local space
+        if sh.colidx.numel():
+            assert int(sh.colidx.max()) < sh.n_ext
+        if sh.t_colidx.numel():
+            assert int(sh.t_colidx.max()) < sh.n_local
+        # every local row's edge count matches the global CSR
+        rp = g.rowptr.numpy()
+        assert np.array_equal(np.diff(sh.rowptr.numpy()),
+                              np.diff(rp)[sh.lo:sh.hi])
+
+    # send plans mirror recv plans: what rank r sends to s is exactly
+    # the slice of s's halo owned by r
+    for r, sh_r in enumerate(shards):
+        off = 0
+        for s, cnt in enumerate(sh_r.send_splits):
+            sent = sh_r.send_idx[off:off + cnt].numpy() + sh_r.lo
+            off += cnt
+            sh_s = shards[s]
+            ho = 0
+            want = np.empty(0, dtype=np.int64)
+            for o, rc in enumerate(sh_s.recv_splits):
+                if o == r:
+                    want = sh_s.halo_ids[ho:ho + rc].numpy()
+                ho += rc
+            assert np.array_equal(np.sort(sent), np.sort(want)), (r, s)
