@@ -27,6 +27,11 @@ def main():
     from roc_amd.graph import synthetic_graph
     from roc_amd import _C
 
+    if not torch.cuda.is_available():
+        print("bench_vs_libs: GPU-only A/B (torch->rocSPARSE/hipBLASLt "
+              "vs roc_amd kernels); run on an MI355X box")
+        return
+
     dev = "cuda:0"
     print("== SpMM: ours vs torch.sparse (rocSPARSE), fp32, Reddit shape ==")
     g = synthetic_graph(232965, 114848857, seed=1)
