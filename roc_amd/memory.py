@@ -147,8 +147,9 @@ class ActivationOffload:
         self._release(entry["cpu"])
         entry["cpu"] = None
         entry["gpu"] = None
-        if entry in self._entries:
-            self._entries.remove(entry)
+        # identity-based removal: dict equality would compare the tensors
+        # held by OTHER entries elementwise (ambiguous-bool RuntimeError)
+        self._entries = [e for e in self._entries if e is not entry]
         self._start_next()                    # keep the pipeline full
         return gpu
 
