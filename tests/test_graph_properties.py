@@ -26,7 +26,7 @@ def _degree_multiset(g):
     return sorted(np.diff(g.rowptr.numpy()).tolist())
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(graphs())
 def test_transpose_involution(g):
     # (A^T)^T == A as a multigraph: edge multiset preserved
@@ -40,7 +40,7 @@ def test_transpose_involution(g):
     assert int(trp[-1]) == g.num_edges
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(graphs())
 def test_orderings_are_permutations_preserving_structure(g):
     for fn in (degree_order, rcm_order, cluster_order):
@@ -61,7 +61,7 @@ def test_orderings_are_permutations_preserving_structure(g):
         assert e1 == e2, fn.__name__
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(st.integers(2, 60), st.integers(0, 2**31 - 1), st.integers(1, 6))
 def test_edge_balanced_bounds_cover(n, seed, parts):
     from roc_amd.parallel.partition import edge_balanced_bounds
@@ -71,7 +71,7 @@ def test_edge_balanced_bounds_cover(n, seed, parts):
     assert all(b[i] <= b[i + 1] for i in range(len(b) - 1))
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(graphs(), st.integers(1, 5))
 def test_halo_plan_invariants(g, world):
     """Structural contracts of the per-rank halo plan for ANY graph and
