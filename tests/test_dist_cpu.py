@@ -252,3 +252,66 @@ def test_sharded_training_ws3(mode):
     # so cross-world-size agreement is approximate; ranks agree to 1e-6
     assert np.allclose(res[0][1], model.weights[0].detach().numpy(),
                        atol=2e-3)
+
+
+def _elastic_worker(rank, port, ckpt, q):
+    # resume a ws=1 checkpoint in a ws=2 job (weights/Adam state are
+    # replicated, so restarts may change the GPU count freely)
+    try:
+        _init(rank, port)
+        torch.manual_seed(0)
+        g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.05,
+                                                      seed=3)
+        bounds = edge_balanced_bounds(g.rowptr, WS)
+        sh = build_shard(g, rank, WS, bounds)
+        model = build_model("gcn", [feats.shape[1], 16, c], dropout=0.0,
+                            seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
+        tr = Trainer(model, sh, feats, labels, mask, opt)
+        from roc_amd.utils import load_checkpoint
+        load_checkpoint(ckpt, tr)
+        for _ in range(2):
+            tr.train_epoch()
+        q.put((rank, model.weights[0].detach().numpy().copy(), None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, None, repr(e)))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_elastic_resume_ws1_to_ws2(tmp_path):
+    from roc_amd.utils import save_checkpoint, load_checkpoint
+
+    def fresh():
+        torch.manual_seed(0)
+        g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.05,
+                                                      seed=3)
+        sh = build_shard(g, 0, 1)
+        model = build_model("gcn", [feats.shape[1], 16, c], dropout=0.0,
+                            seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
+        return Trainer(model, sh, feats, labels, mask, opt)
+
+    tr = fresh()
+    for _ in range(3):
+        tr.train_epoch()
+    ckpt = str(tmp_path / "elastic.pt")
+    save_checkpoint(ckpt, tr)
+    for _ in range(2):  # ws=1 continuation = the oracle
+        tr.train_epoch()
+    w_ref = tr.model.weights[0].detach().numpy()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_elastic_worker, args=(r, 29531, ckpt, q))
+             for r in range(WS)]
+    for p in procs:
+        p.start()
+    res = sorted([q.get() for _ in range(WS)], key=lambda t: t[0])
+    for p in procs:
+        p.join(timeout=180)
+    for rank, w, err in res:
+        assert err is None, f"rank {rank}: {err}"
+    assert np.allclose(res[0][1], res[1][1], atol=1e-6)
+    assert np.allclose(res[0][1], w_ref, atol=2e-3)  # Adam noise tolerance
