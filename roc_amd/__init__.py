@@ -13,7 +13,9 @@ designed for AMD Instinct MI355X (gfx950/CDNA4):
 __version__ = "0.1.0"
 
 from . import graph  # noqa: F401
-from .graph import CSRGraph, load_lux, synthetic_graph, synthetic_dataset  # noqa: F401
+from .graph import (CSRGraph, load_lux, synthetic_graph,  # noqa: F401
+                    synthetic_dataset, reorder_graph, apply_ordering,
+                    ORDERINGS)
 from .parallel.partition import GraphShard, build_shard, edge_balanced_bounds  # noqa: F401
 from .optim import AdamOptimizer  # noqa: F401
 from .engine import Trainer  # noqa: F401
