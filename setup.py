@@ -17,7 +17,10 @@ from torch.utils.cpp_extension import CUDAExtension, BuildExtension
 
 def sources():
     srcs = sorted(glob.glob("roc_amd/csrc/*.cpp"))
-    srcs += sorted(glob.glob("roc_amd/csrc/*.hip"))
+    # exclude torch-hipify's generated *_hip.hip copies from earlier
+    # builds (gitignored, but present in a built tree)
+    srcs += sorted(s for s in glob.glob("roc_amd/csrc/*.hip")
+                   if not s.endswith("_hip.hip"))
     return srcs
 
 
