@@ -38,8 +38,12 @@ def pad_features(feats: torch.Tensor, mult: int = 8) -> torch.Tensor:
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--steps", type=int, default=100)
     ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--exact-steps", action="store_true",
+                    help="never auto-extend the timed region (default: "
+                         "extend --steps so the timed region is >=2s, so "
+                         "SMI sampling can observe the busy GPU)")
     ap.add_argument("--dataset", default="reddit")
     ap.add_argument("--model", default="gcn")
     ap.add_argument("--hidden", type=int, default=256)
@@ -145,11 +149,28 @@ def main():
     # at least 4 warmup epochs so hipGraph capture (2 warmups + capture)
     # never lands inside the timed region
     warmup_done = max(args.warmup, 4 if trainer.use_graph else args.warmup)
+    barrier()
+    tw = time.perf_counter()
     for _ in range(warmup_done):
         trainer.train_epoch()
     barrier()
+    est = (time.perf_counter() - tw) / warmup_done
+    steps = args.steps
+    if not args.exact_steps and est * steps < 2.0:
+        # extend the timed region to >=2s so the driver's SMI sampling
+        # observes a busy GPU; every rank must agree on the step count
+        steps = min(max(steps, int(2.5 / max(est, 1e-6)) + 1), 5000)
+        if world > 1:
+            t = torch.tensor([steps], dtype=torch.int64,
+                             device=device if on_gpu else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            steps = int(t.item())
+        if rank == 0 and steps != args.steps:
+            print(f"[bench] timed region extended: --steps {args.steps} "
+                  f"-> {steps} (~{est*1e3:.1f} ms/epoch; pass "
+                  f"--exact-steps to disable)", file=sys.stderr, flush=True)
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for _ in range(steps):
         trainer.train_epoch()
     barrier()
     elapsed = time.perf_counter() - t0
@@ -160,7 +181,7 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    per_epoch = elapsed / args.steps
+    per_epoch = elapsed / steps
     if rank == 0:
         metric_name = ("per-epoch training time (s), 2-layer GCN on Reddit"
                        if (args.model, args.layers, args.dataset)
@@ -172,7 +193,7 @@ def main():
             "value": per_epoch,
             "unit": "s/epoch",
             "n_gpus": world,
-            "steps": args.steps,
+            "steps": steps,
             "warmup": warmup_done,
             "ms_per_step": per_epoch * 1e3,
             "higher_is_better": False,

@@ -24,6 +24,7 @@ void adam_step(torch::Tensor w, torch::Tensor g, torch::Tensor m,
 void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu,
              c10::optional<torch::Tensor> row_scale);
 void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B);
+void spmm_refresh_knobs();
 void register_graph_cpu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -57,5 +58,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("C"), pybind11::arg("A"), pybind11::arg("Bt"),
         pybind11::arg("relu"), pybind11::arg("row_scale") = pybind11::none());
   m.def("gemm_atb", &gemm_atb, "C += A^T @ B (fp32 split-K accumulate)");
+  m.def("spmm_refresh_knobs", &spmm_refresh_knobs,
+        "re-read ROC_SPMM_* env knobs (A/B harness only)");
   register_graph_cpu(m);
 }

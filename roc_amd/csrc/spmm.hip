@@ -262,29 +262,55 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
   }
 }
 
+// A/B geometry knobs, pinned ONCE at first launch (a per-launch getenv
+// on the hot path is a footgun under hipGraph capture: replays would
+// silently keep whatever value was live at capture; now ALL launches —
+// captured or eager — use the same process-lifetime values).
+struct SpmmKnobs {
+  int team_override;  // 0 = auto (by D); else 8/16/32/64
+  int unroll;         // 4 | 8 | 16 (16 measured best)
+  bool allow_buffer;  // SRSRC buffer_load gather path
+};
+
+static SpmmKnobs read_spmm_knobs() {
+  SpmmKnobs k{0, 16, true};
+  if (const char* e = getenv("ROC_SPMM_TEAM")) {
+    const int t = atoi(e);
+    if (t == 8 || t == 16 || t == 32 || t == 64) k.team_override = t;
+  }
+  if (const char* e = getenv("ROC_SPMM_UNROLL")) k.unroll = atoi(e);
+  if (const char* e = getenv("ROC_SPMM_BUFFER")) k.allow_buffer = e[0] != '0';
+  return k;
+}
+
+static SpmmKnobs g_spmm_knobs;
+static bool g_spmm_knobs_read = false;
+
+static const SpmmKnobs& spmm_knobs() {
+  if (!g_spmm_knobs_read) {
+    g_spmm_knobs = read_spmm_knobs();
+    g_spmm_knobs_read = true;
+  }
+  return g_spmm_knobs;
+}
+
 template <typename T>
 void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
                  const float* deg_dst, const float* deg_src,
                  const int* row_order, int num_rows, int64_t D,
                  bool accumulate, size_t x_elems, hipStream_t stream) {
   constexpr int EPU = EltTraits<T>::kPerVec;
+  const SpmmKnobs& kn = spmm_knobs();
   const int64_t units = (D + EPU - 1) / EPU;
   int team = 8;
   while (team < units && team < 64) team *= 2;
-  const char* team_env = getenv("ROC_SPMM_TEAM");  // geometry experiments
-  if (team_env) {
-    const int t = atoi(team_env);
-    if (t == 8 || t == 16 || t == 32 || t == 64) team = t;
-  }
+  if (kn.team_override) team = kn.team_override;
   const int col_tiles = (int)((units + team - 1) / team);
   const int tpb = kBlock / team;
   dim3 grid(roc_grid_1d(num_rows, tpb, 8192), col_tiles);
-  // read per call (cheap vs a ms-scale launch) so A/B harnesses can toggle
-  const char* un_env = getenv("ROC_SPMM_UNROLL");
-  const int un = un_env ? atoi(un_env) : 16;  // 4 | 8 | 16 (16 measured best)
-  const char* buf_env = getenv("ROC_SPMM_BUFFER");
+  const int un = kn.unroll;
   const size_t xb = x_elems * sizeof(T);
-  const bool buf = xb < (size_t)UINT_MAX && !(buf_env && buf_env[0] == '0');
+  const bool buf = xb < (size_t)UINT_MAX && kn.allow_buffer;
   const unsigned x_bytes = (unsigned)(buf ? xb : 0);
   // measured (scripts/bench_spmm.py, Reddit shape): degree-descending
   // scheduling wins for wide rows (D=256: -13%) but loses for narrow
@@ -327,6 +353,10 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
 }
 
 }  // namespace
+
+// A/B harness hook (scripts/bench_spmm.py): re-read the geometry env
+// vars. Never call while a hipGraph capture of spmm launches is live.
+void spmm_refresh_knobs() { g_spmm_knobs_read = false; }
 
 void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
           torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,

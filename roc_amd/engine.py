@@ -222,30 +222,50 @@ class Trainer:
 
     # -- cost-model repartitioning (the MLSys'20 Roc idea; the reference
     #    code only has the static edge-balanced split) --------------------
-    def measure_and_rebalance(self, feats, labels, mask, probe_epochs=3):
-        """Measure this rank's epoch time, gather all ranks' times, re-split
-        the vertex range so predicted load matches measured throughput, and
-        rebuild the shard in place. Needs the FULL feats/labels/mask (every
-        rank loads/generates the full dataset at startup).
+    def measure_and_rebalance(self, feats=None, labels=None, mask=None,
+                              probe_epochs=3):
+        """Measure this rank's epoch time, gather every rank's (time,
+        halo-rows) over the gloo control plane, fit the comm-aware cost
+        model t ≈ a*edges + b*halo, re-split the vertex ranges so
+        predicted times equalize, and rebuild the shard in place.
+
+        Two data modes: full (attach_full_graph + full feats/labels/mask
+        on every rank) or windowed (attach_windowed_dataset — each rank
+        re-reads only its new window from the dataset files).
         Returns the new bounds."""
-        from .parallel.partition import rebalance_bounds, build_shard
+        from .parallel.comm import cpu_group
+        from .parallel.partition import (rebalance_bounds_comm, build_shard,
+                                         build_shard_from_lux)
         t = self.timed_epochs(probe_epochs) / probe_epochs
         if self.shard.world_size == 1 or not dist.is_initialized():
             return self.shard.bounds
-        dev = self.device if self.device.type == "cuda" else "cpu"
-        times = torch.zeros(self.shard.world_size, device=dev)
-        times[self.shard.rank] = t
-        dist.all_reduce(times, group=self.group)
-        times = times.cpu()
-        # rowptr of the full graph is not kept; derive per-rank edge counts
-        # from the current shard (identical on all ranks via bounds)
-        new_bounds = rebalance_bounds(
-            self._full_rowptr, self.shard.bounds, times.tolist())
+        ws = self.shard.world_size
+        g = cpu_group(self.group)
+        stats = torch.zeros(ws, 2, dtype=torch.float64)
+        stats[self.shard.rank, 0] = t
+        stats[self.shard.rank, 1] = float(self.shard.n_halo)
+        dist.all_reduce(stats, group=g)
+        if not hasattr(self, "_rebal_samples"):
+            self._rebal_samples = []
+        new_bounds = rebalance_bounds_comm(
+            self._full_rowptr, self.shard.bounds, stats[:, 0].tolist(),
+            stats[:, 1].tolist(), self._rebal_samples)
         if new_bounds == self.shard.bounds:
             return new_bounds
-        new_shard = build_shard(self._full_graph, self.shard.rank,
-                                self.shard.world_size, new_bounds)
-        self.load_shard(new_shard, feats, labels, mask)
+        if getattr(self, "_window_loader", None) is not None:
+            new_shard = build_shard_from_lux(
+                self._lux_path, self.shard.rank, ws, new_bounds,
+                group=self.group)
+            feats, labels, mask = self._window_loader(
+                new_shard.lo, new_shard.hi)
+            self.load_shard(new_shard, feats, labels, mask,
+                            local_slices=True)
+        else:
+            new_shard = build_shard(self._full_graph, self.shard.rank,
+                                    ws, new_bounds,
+                                    use_comm=dist.is_initialized(),
+                                    group=self.group)
+            self.load_shard(new_shard, feats, labels, mask)
         return new_bounds
 
     def attach_full_graph(self, g):
@@ -253,10 +273,20 @@ class Trainer:
         self._full_graph = g
         self._full_rowptr = g.rowptr
 
-    def load_shard(self, shard, feats, labels, mask):
+    def attach_windowed_dataset(self, lux_path: str, window_loader,
+                                rowptr: torch.Tensor):
+        """Repartitioning under windowed file loading: `window_loader(lo,
+        hi)` returns this rank's (feats, labels, mask) rows; `rowptr` is
+        the full-graph row pointer (cheap: 8 B/vertex, already read by
+        load_lux_meta)."""
+        self._lux_path = lux_path
+        self._window_loader = window_loader
+        self._full_rowptr = rowptr
+
+    def load_shard(self, shard, feats, labels, mask, local_slices=False):
         device, dt = self.device, self.dtype
         self.shard = shard.to(device)
-        lo, hi = shard.lo, shard.hi
+        lo, hi = (0, shard.n_local) if local_slices else (shard.lo, shard.hi)
         self.x = feats[lo:hi].to(device=device, dtype=dt).contiguous()
         self.labels = labels[lo:hi].to(device=device).contiguous()
         self.mask = mask[lo:hi].to(device=device,

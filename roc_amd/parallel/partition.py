@@ -127,6 +127,55 @@ def edge_balanced_bounds(rowptr: torch.Tensor, num_parts: int) -> List[int]:
     return bounds
 
 
+def fit_cost_model(samples: List[tuple]) -> tuple:
+    """Least-squares fit of per-rank epoch time ≈ a*edges + b*halo_rows
+    over measured (edges, halo, time) samples (accumulated across
+    rebalance rounds — the MLSys'20 online-regression partitioner idea).
+    Falls back to the pure-edge model (b=0) when the system is
+    underdetermined or the comm coefficient comes out negative."""
+    A = np.array([[s[0], s[1]] for s in samples], dtype=np.float64)
+    y = np.array([s[2] for s in samples], dtype=np.float64)
+    a = float((A[:, 0] * y).sum() / max((A[:, 0] ** 2).sum(), 1e-30))
+    b = 0.0
+    if len(samples) >= 3 and np.linalg.matrix_rank(A) == 2:
+        coef, *_ = np.linalg.lstsq(A, y, rcond=None)
+        if coef[0] > 0 and coef[1] >= 0:
+            a, b = float(coef[0]), float(coef[1])
+    return a, b
+
+
+def rebalance_bounds_comm(rowptr: torch.Tensor, bounds: List[int],
+                          per_rank_time: List[float],
+                          per_rank_halo: List[int],
+                          samples: Optional[List[tuple]] = None) -> List[int]:
+    """Comm-aware refinement: fit t_k ≈ a*edges_k + b*halo_k, then give
+    each rank an edge budget e'_k = (T - b*halo_k)/a so PREDICTED totals
+    equalize (halo_k of the new range approximated by the current one —
+    a fixed point the repeated rebalance iterates toward). `samples`
+    (mutated in place) carries measurement history across calls."""
+    rp = rowptr.numpy().astype(np.float64)
+    p = len(bounds) - 1
+    edges = [rp[bounds[k + 1]] - rp[bounds[k]] for k in range(p)]
+    cur = [(edges[k], float(per_rank_halo[k]), per_rank_time[k])
+           for k in range(p)]
+    if samples is not None:
+        samples.extend(cur)
+        cur = samples[-8 * p:]  # bounded history window
+    a, b = fit_cost_model(cur)
+    T = sum(per_rank_time) / p
+    e_total = rp[-1]
+    floor = 0.05 * e_total / p
+    budget = [max((T - b * per_rank_halo[k]) / max(a, 1e-30), floor)
+              for k in range(p)]
+    scale = e_total / sum(budget)
+    targets = np.cumsum([0.0] + [bk * scale for bk in budget])
+    nb = np.searchsorted(rp, targets, side="left").tolist()
+    nb[0], nb[-1] = 0, rowptr.numel() - 1
+    for i in range(1, len(nb)):
+        nb[i] = max(nb[i], nb[i - 1])
+    return nb
+
+
 def rebalance_bounds(rowptr: torch.Tensor, bounds: List[int],
                      per_rank_time: List[float]) -> List[int]:
     """Cost-model refinement (the MLSys'20 Roc idea the reference code lacks):
@@ -150,8 +199,14 @@ def _send_plan_comm(halo_ids: np.ndarray, recv_splits: List[int], lo: int,
                     world_size: int, group):
     """Exchange halo REQUESTS over the process group: each rank learns
     which of its rows the others need without scanning their edge
-    windows (required for windowed dataset loading at scale)."""
+    windows (required for windowed dataset loading at scale). The
+    request tensors are CPU int64, so this rides the gloo control
+    plane (cpu_group) — the training group is typically RCCL, which
+    cannot carry CPU tensors."""
     import torch.distributed as dist
+
+    from .comm import cpu_group
+    group = cpu_group(group)
     counts_in = torch.tensor(recv_splits, dtype=torch.int64)
     counts_out = torch.empty(world_size, dtype=torch.int64)
     dist.all_to_all_single(counts_out, counts_in, group=group)

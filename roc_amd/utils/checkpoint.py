@@ -56,6 +56,12 @@ def load_checkpoint(path: str, trainer) -> dict:
     counter = F._DROPOUT_STATE.get("counter")
     F._DROPOUT_STATE.update(state["dropout_state"])
     F._DROPOUT_STATE["counter"] = counter  # device tensor is not persisted
+    # the saved seed is RANK 0's (= the base seed); re-derive this
+    # rank's decorrelated seed exactly as Trainer.__init__ does, so a
+    # resumed multi-rank job keeps distinct per-rank dropout streams
+    # (rank 0 resume stays bit-exact: base + 0*7919 == base)
+    F._DROPOUT_STATE["seed"] = (state["dropout_state"]["seed"]
+                                + trainer.shard.rank * 7919)
     torch.set_rng_state(state["torch_rng"])
     F.bump_weight_version()  # invalidate cached weight casts (in-place load)
     return state.get("extra", {})

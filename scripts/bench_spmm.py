@@ -66,6 +66,7 @@ def main():
         for un in (8, 16):
             for use_order in (0, 1):
                 os.environ["ROC_SPMM_UNROLL"] = str(un)
+                _C.spmm_refresh_knobs()  # knobs are pinned at first launch
                 ro = row_order if use_order else None
 
                 def fn(ro=ro):
@@ -77,6 +78,7 @@ def main():
                 print(f"  {key}: median {med:8.2f} ms  best {best:8.2f}"
                       f"  ({gb/med:.2f} TB/s eff)", flush=True)
         os.environ.pop("ROC_SPMM_UNROLL", None)
+        _C.spmm_refresh_knobs()
     print("done")
 
 

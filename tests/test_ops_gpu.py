@@ -264,6 +264,30 @@ def test_softmax_ce_gpu(dtype):
         md_want["ce_loss"], rel=3e-2)
 
 
+@pytest.mark.parametrize("c", [107, 172])
+def test_softmax_ce_wide_unpadded_gpu(c):
+    """C > 64 takes the wide path (softmax_ce.hip multi-pass); run it
+    UNPADDED at the amazon (107) and papers (172) class counts against
+    the fp32 reference — fwd metrics AND backward grad."""
+    torch.manual_seed(21)
+    n = 1500
+    logits = torch.randn(n, c, device=DEV, requires_grad=True)
+    labels = torch.randint(0, c, (n,), device=DEV)
+    mask = torch.randint(1, 4, (n,), dtype=torch.int32, device=DEV)
+    loss, metrics = F.softmax_cross_entropy(logits, labels, mask)
+    loss.backward()
+    md = F.decode_metrics(metrics)
+    lc = logits.detach().cpu().requires_grad_(True)
+    dl_want, md_want = ref.softmax_cross_entropy(
+        lc.detach().float(), labels.cpu(), mask.cpu())
+    assert md["train_total"] == md_want["train_total"]
+    assert md["train_acc"] == pytest.approx(
+        md_want["train_correct"] / md_want["train_total"], abs=1e-3)
+    assert md["ce_loss"] == pytest.approx(md_want["ce_loss"], rel=3e-2)
+    assert torch.allclose(logits.grad.cpu(), dl_want, atol=1e-4), \
+        (logits.grad.cpu() - dl_want).abs().max()
+
+
 def test_softmax_ce_grad_gpu():
     torch.manual_seed(10)
     n, c = 500, 41

@@ -208,6 +208,12 @@ def main():
     if args.dtype == "bf16" or (args.dtype == "auto" and on_gpu):
         dtype = torch.bfloat16
     n_train = int((mask == 1).sum())
+    if world > 1 and dist.is_initialized() and pre_shard is not None:
+        # windowed loading: mask is the LOCAL slice, so sum the train
+        # counts so every rank uses the GLOBAL 1/n_train grad scale
+        # (otherwise the summed all-reduced gradient is rank-weighted)
+        from roc_amd.parallel.comm import allreduce_scalar_int
+        n_train = allreduce_scalar_int(n_train)
     gs = 1.0 if args.loss == "sum" else 1.0 / max(n_train, 1)
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
                       compute_dtype=dtype, grad_scale=gs, seed=args.seed,
@@ -232,7 +238,26 @@ def main():
               f"dropout={args.dropout} comm={shard.comm_mode}", flush=True)
 
     if args.rebalance_every:
-        trainer.attach_full_graph(g)
+        if g is not None:
+            trainer.attach_full_graph(g)
+        else:
+            # windowed loading: rebalance re-reads only the new window
+            from roc_amd.graph import (load_lux_meta, load_features_window)
+            lux = args.file + ".add_self_edge.lux"
+            n = load_lux_meta(lux)[0]
+            raw_dim = int(args.layers.split("-")[0])  # on-disk width
+            padded_dim = feats.shape[1]
+            full_labels = load_labels(args.file + ".label", n)
+            full_mask = load_mask(args.file + ".mask", n)
+
+            def _loader(lo, hi, _n=n, _d=raw_dim, _p=padded_dim):
+                f = load_features_window(args.file, _n, _d, lo, hi)
+                if f.shape[1] < _p:
+                    f = torch.nn.functional.pad(f, (0, _p - f.shape[1]))
+                return f, full_labels[lo:hi], full_mask[lo:hi]
+
+            trainer.attach_windowed_dataset(lux, _loader,
+                                            load_lux_meta(lux)[2])
 
     t_start = time.perf_counter()
     while trainer.epoch < args.epochs:
@@ -240,13 +265,25 @@ def main():
         ep = trainer.epoch
         if args.auto_recover and (args.eval_every == 0
                                   or ep % max(args.eval_every, 1) == 0):
+            # the divergence flag is agreed via MAX all-reduce so every
+            # rank takes the restore branch together (a NaN confined to
+            # one rank's rows must not desynchronize collective counts)
+            err = None
             try:
                 check_metrics(metrics)
             except TrainingDiverged as e:
+                err = e
+            bad = int(err is not None)
+            if world > 1 and dist.is_initialized():
+                from roc_amd.parallel.comm import allreduce_scalar_int
+                bad = allreduce_scalar_int(bad, op="max")
+            if bad:
                 if not (args.checkpoint and os.path.exists(args.checkpoint)):
-                    raise
+                    raise err or TrainingDiverged(
+                        "divergence detected on a peer rank")
                 if rank == 0:
-                    print(f"[recover] {e}; restoring {args.checkpoint}, "
+                    print(f"[recover] {err or 'peer-rank divergence'}; "
+                          f"restoring {args.checkpoint}, "
                           f"lr {trainer.optimizer.lr} -> "
                           f"{trainer.optimizer.lr * 0.5}", flush=True)
                 load_checkpoint(args.checkpoint, trainer)
