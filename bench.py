@@ -65,9 +65,14 @@ def main():
         dist.init_process_group("nccl" if on_gpu else "gloo",
                                 rank=rank, world_size=world)
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
-        # ROC_DEVICE_OVERRIDE: testing hook (e.g. 2 RCCL ranks on 1 GPU)
+        # ROC_DEVICE_OVERRIDE: testing hook. NOTE: RCCL refuses two
+        # ranks on ONE device (measured: "Duplicate GPU detected",
+        # gpurun_out/r2c1); for single-box multi-rank validation use
+        # CPX compute partitioning instead (scripts/gpu_r2_call2.sh)
+        # and the natural local_rank -> device mapping below.
         dev_idx = int(os.environ.get("ROC_DEVICE_OVERRIDE", local_rank))
         if on_gpu:
+            dev_idx %= max(torch.cuda.device_count(), 1)
             torch.cuda.set_device(dev_idx)
             device = f"cuda:{dev_idx}"
         else:

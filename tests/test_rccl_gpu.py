@@ -5,13 +5,16 @@ reduce-to-owner, flat-grad all-reduce, metrics all-reduce) were only
 ever executed over gloo on CPU in round 1; the driver's round-end
 8-GPU scale run would have been the first-ever RCCL execution. These
 tests run every one of those communication paths on the `nccl`
-backend (= RCCL on ROCm) with world_size=2, BOTH ranks pinned to the
-one leased MI355X — the ROC_DEVICE_OVERRIDE trick bench.py supports —
+backend (= RCCL on ROCm) with world_size=2 on the one leased MI355X
 and assert equality with the single-rank GPU run.
 
-If this RCCL build refuses two ranks on one device ("Duplicate GPU
-detected"), the tests SKIP with that reason recorded — that outcome
-still tells us the 8-GPU path has to be validated rank-per-device.
+RCCL 2.26.6 hard-refuses two ranks on one device ("Duplicate GPU
+detected", measured gpurun_out/r2c1_ws2_halo_eager.log), so the
+harness first needs the GPU split into logical devices via CPX
+compute partitioning (`amd-smi set --compute-partition CPX`, see
+scripts/gpu_r2_call2.sh); ranks then map rank -> cuda:(rank %
+device_count). On an unpartitioned (SPX) box the duplicate-GPU error
+is converted to a SKIP with the reason recorded.
 """
 import datetime
 import os
@@ -27,6 +30,14 @@ WS = 2
 _UNSUPPORTED_MARKERS = ("duplicate gpu", "invalid usage", "invalidusage")
 
 
+def _device(rank) -> str:
+    # CPX compute partitioning splits the one MI355X into multiple
+    # logical devices -> real distinct-device RCCL; in SPX (1 device)
+    # both ranks pin device 0 and RCCL's duplicate-GPU check fires,
+    # which the harness converts to a SKIP with the reason recorded
+    return f"cuda:{rank % torch.cuda.device_count()}"
+
+
 def _init(rank, port, comm_mode="halo"):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -35,7 +46,7 @@ def _init(rank, port, comm_mode="halo"):
     # after the timeout when blocking-wait is on
     os.environ["TORCH_NCCL_BLOCKING_WAIT"] = "1"
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    torch.cuda.set_device(0)  # both ranks share the one GPU
+    torch.cuda.set_device(torch.device(_device(rank)))
     dist.init_process_group(
         "nccl", rank=rank, world_size=WS,
         timeout=datetime.timedelta(seconds=120))
@@ -75,7 +86,7 @@ def _run(worker, port, timeout=300):
 def _collectives_worker(rank, port, q):
     try:
         _init(rank, port)
-        dev = torch.device("cuda:0")
+        dev = torch.device(_device(rank))
         out = {}
         # flat-grad all-reduce (fp32)
         t = torch.full((1024,), float(rank + 1), device=dev)
@@ -150,7 +161,7 @@ def _train_worker(rank, port, q):
         _init(rank, port)
         out = {}
         for mode in ("halo", "allgather"):
-            md, w0 = _train("cuda:0", rank, WS, mode)
+            md, w0 = _train(_device(rank), rank, WS, mode)
             out[mode] = (md, w0)
         q.put((rank, out, None))
     except Exception as e:  # noqa: BLE001
