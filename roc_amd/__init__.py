@@ -16,7 +16,8 @@ from . import graph  # noqa: F401
 from .graph import (CSRGraph, load_lux, synthetic_graph,  # noqa: F401
                     synthetic_dataset, reorder_graph, apply_ordering,
                     ORDERINGS)
-from .parallel.partition import GraphShard, build_shard, edge_balanced_bounds  # noqa: F401
+from .parallel.partition import (GraphShard, build_shard,  # noqa: F401
+                                 edge_balanced_bounds, edge_tensor)
 from .optim import AdamOptimizer  # noqa: F401
 from .engine import Trainer  # noqa: F401
 from .models import build_model, GCN, GraphSAGE, GIN  # noqa: F401

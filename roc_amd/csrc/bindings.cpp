@@ -25,6 +25,12 @@ void gemm_rr(torch::Tensor C, torch::Tensor A, torch::Tensor Bt, bool relu,
              c10::optional<torch::Tensor> row_scale);
 void gemm_atb(torch::Tensor C, torch::Tensor A, torch::Tensor B);
 void spmm_refresh_knobs();
+void spmm_edge(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
+               torch::Tensor colidx, torch::Tensor edge_val,
+               c10::optional<torch::Tensor> deg_dst,
+               c10::optional<torch::Tensor> row_order, bool accumulate);
+void edge_dot(torch::Tensor dw, torch::Tensor dy, torch::Tensor x,
+              torch::Tensor rowptr, torch::Tensor colidx);
 void register_graph_cpu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -60,5 +66,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_atb", &gemm_atb, "C += A^T @ B (fp32 split-K accumulate)");
   m.def("spmm_refresh_knobs", &spmm_refresh_knobs,
         "re-read ROC_SPMM_* env knobs (A/B harness only)");
+  m.def("spmm_edge", &spmm_edge,
+        "per-edge-weighted CSR SpMM (edge-tensor consumer)",
+        pybind11::arg("out"), pybind11::arg("x"), pybind11::arg("rowptr"),
+        pybind11::arg("colidx"), pybind11::arg("edge_val"),
+        pybind11::arg("deg_dst") = pybind11::none(),
+        pybind11::arg("row_order") = pybind11::none(),
+        pybind11::arg("accumulate") = false);
+  m.def("edge_dot", &edge_dot,
+        "dw[e] = <dy[row_e], x[col_e]> (edge-value gradient)");
   register_graph_cpu(m);
 }

@@ -352,6 +352,204 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
 #undef ROC_SPMM_L3
 }
 
+// ---------------------------------------------------------------------------
+// Edge-weighted SpMM + its edge-value gradient (the edge-tensor ops).
+// out[v,:] (+)= deg_dst[v] * sum_{e in row v} edge_val[e] * x[col_e,:]
+// Same row-team geometry as spmm_kernel; weights are per-EDGE (not
+// per-source like deg_src), loaded sequentially -> one scalar broadcast
+// per gather. 4-deep unrolled; capability path, not the fused-GCN hot
+// path, so one unroll depth is enough.
+// ---------------------------------------------------------------------------
+
+// per-row accumulate loop shared by both gather engines (8 gathers in
+// flight via 2x 4-group issue; per-edge scalar weight broadcast)
+template <typename T, int EPU, typename LD>
+__device__ __forceinline__ void edge_accum_row(
+    float* __restrict__ acc, const LD& ld, const int* __restrict__ colidx,
+    const float* __restrict__ edge_val, int64_t e, int64_t e1) {
+  using Raw = typename RawVec<T>::type;
+  for (; e + 3 < e1; e += 4) {
+    const int u0 = colidx[e], u1 = colidx[e + 1];
+    const int u2 = colidx[e + 2], u3 = colidx[e + 3];
+    const float w0 = edge_val[e], w1 = edge_val[e + 1];
+    const float w2 = edge_val[e + 2], w3 = edge_val[e + 3];
+    const Raw r0 = ld.load(u0), r1 = ld.load(u1);
+    const Raw r2 = ld.load(u2), r3 = ld.load(u3);
+    acc_add(acc, r0, w0); acc_add(acc, r1, w1);
+    acc_add(acc, r2, w2); acc_add(acc, r3, w3);
+  }
+  for (; e < e1; ++e) {
+    const Raw r0 = ld.load(colidx[e]);
+    acc_add(acc, r0, edge_val[e]);
+  }
+}
+
+template <typename T, int TEAM, bool BUF>
+__global__ __launch_bounds__(kBlock) void spmm_edge_kernel(
+    T* __restrict__ out, const T* __restrict__ x,
+    const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
+    const float* __restrict__ edge_val, const float* __restrict__ deg_dst,
+    const int* __restrict__ row_order, int num_rows, int64_t D,
+    bool accumulate, unsigned x_bytes) {
+  constexpr int EPU = EltTraits<T>::kPerVec;
+  const int tpb = kBlock / TEAM;
+  const int team = blockIdx.x * tpb + (int)threadIdx.x / TEAM;
+  const int lane = (int)threadIdx.x % TEAM;
+  const int nteams = gridDim.x * tpb;
+  const int64_t col0 = ((int64_t)blockIdx.y * TEAM + lane) * EPU;
+  const bool full = (col0 + EPU) <= D;
+  const int nvalid = full ? EPU : (col0 < D ? (int)(D - col0) : 0);
+
+  if (full) {
+    for (int ri = team; ri < num_rows; ri += nteams) {
+      const int row = row_order ? row_order[ri] : ri;
+      const int64_t e0 = rowptr[row];
+      const int64_t e1 = rowptr[row + 1];
+      float acc[EPU];
+      T* o = out + (int64_t)row * D + col0;
+      if (accumulate) {
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) acc[j] = elt_to_f32(o[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
+      }
+      if constexpr (BUF) {
+        BufferGather<T> ld{
+            __builtin_amdgcn_make_buffer_rsrc((void*)x, (short)0, x_bytes,
+                                              0x00020000),
+            (unsigned)(D * sizeof(T)), (unsigned)(col0 * sizeof(T))};
+        edge_accum_row<T, EPU>(acc, ld, colidx, edge_val, e0, e1);
+      } else {
+        GlobalGather<T> ld{x, D, col0};
+        edge_accum_row<T, EPU>(acc, ld, colidx, edge_val, e0, e1);
+      }
+      if (deg_dst) {
+        const float s = deg_dst[row];
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) acc[j] *= s;
+      }
+      if constexpr (EPU == 8) store_bf16x8(o, acc); else store_f32x4(o, acc);
+    }
+  } else if (nvalid > 0) {
+    for (int ri = team; ri < num_rows; ri += nteams) {
+      const int row = row_order ? row_order[ri] : ri;
+      float acc[EPU];
+      T* o = out + (int64_t)row * D + col0;
+      for (int j = 0; j < nvalid; ++j)
+        acc[j] = accumulate ? elt_to_f32(o[j]) : 0.f;
+      for (int64_t e = rowptr[row]; e < rowptr[row + 1]; ++e) {
+        const T* r = x + (int64_t)colidx[e] * D + col0;
+        const float w0 = edge_val[e];
+        for (int j = 0; j < nvalid; ++j) acc[j] += w0 * elt_to_f32(r[j]);
+      }
+      const float s = deg_dst ? deg_dst[row] : 1.f;
+      for (int j = 0; j < nvalid; ++j) f32_to_elt(acc[j] * s, o + j);
+    }
+  }
+}
+
+// dw[e] = <dy[row_e,:], x[col_e,:]>  (fp32 accumulate).
+// Team per row; per edge each lane dots its strided D-chunks, then a
+// log2(TEAM) shfl_xor tree folds the team partials; lane 0 stores.
+// Teams are contiguous, power-of-2-aligned lane groups, so the xor
+// tree never crosses a team boundary within the wave64 front.
+template <typename T, int TEAM>
+__global__ __launch_bounds__(kBlock) void edge_dot_kernel(
+    float* __restrict__ dw, const T* __restrict__ dy,
+    const T* __restrict__ x, const int64_t* __restrict__ rowptr,
+    const int* __restrict__ colidx, int num_rows, int64_t D) {
+  constexpr int EPU = EltTraits<T>::kPerVec;
+  using Raw = typename RawVec<T>::type;
+  const int tpb = kBlock / TEAM;
+  const int team = blockIdx.x * tpb + (int)threadIdx.x / TEAM;
+  const int lane = (int)threadIdx.x % TEAM;
+  const int nteams = gridDim.x * tpb;
+  for (int row = team; row < num_rows; row += nteams) {
+    const int64_t e0 = rowptr[row], e1 = rowptr[row + 1];
+    const T* dyr = dy + (int64_t)row * D;
+    for (int64_t e = e0; e < e1; ++e) {
+      const T* xr = x + (int64_t)colidx[e] * D;
+      float p = 0.f;
+      for (int64_t c = (int64_t)lane * EPU; c + EPU <= D;
+           c += (int64_t)TEAM * EPU) {
+        const Raw a = load_raw(dyr + c);
+        const Raw b = load_raw(xr + c);
+        float ta[EPU] = {0.f}, tb[EPU] = {0.f};
+        acc_add(ta, a, 1.f);
+        acc_add(tb, b, 1.f);
+#pragma unroll
+        for (int j = 0; j < EPU; ++j) p += ta[j] * tb[j];
+      }
+      // scalar tail, lane-strided, for D % EPU != 0 (cold; dims padded)
+      for (int64_t j = (D / EPU) * EPU + lane; j < D; j += TEAM)
+        p += elt_to_f32(dyr[j]) * elt_to_f32(xr[j]);
+#pragma unroll
+      for (int off = TEAM / 2; off > 0; off >>= 1)
+        p += __shfl_xor(p, off, 64);
+      if (lane == 0) dw[e] = p;
+    }
+  }
+}
+
+template <typename T>
+void launch_spmm_edge(T* out, const T* x, const int64_t* rowptr,
+                      const int* colidx, const float* edge_val,
+                      const float* deg_dst, const int* row_order,
+                      int num_rows, int64_t D, bool accumulate,
+                      size_t x_elems, hipStream_t stream) {
+  constexpr int EPU = EltTraits<T>::kPerVec;
+  const int64_t units = (D + EPU - 1) / EPU;
+  int team = 8;
+  while (team < units && team < 64) team *= 2;
+  const int col_tiles = (int)((units + team - 1) / team);
+  const int tpb = kBlock / team;
+  dim3 grid(roc_grid_1d(num_rows, tpb, 8192), col_tiles);
+  const size_t xb = x_elems * sizeof(T);
+  const bool buf = xb < (size_t)UINT_MAX && spmm_knobs().allow_buffer;
+  if (team < 16) row_order = nullptr;
+#define ROC_SPMME_L(TEAM_, BUF_)                                            \
+  hipLaunchKernelGGL((spmm_edge_kernel<T, TEAM_, BUF_>), grid, dim3(kBlock),\
+                     0, stream, out, x, rowptr, colidx, edge_val, deg_dst,  \
+                     row_order, num_rows, D, accumulate,                    \
+                     (unsigned)(buf ? xb : 0))
+#define ROC_SPMME_T(TEAM_)                                                  \
+  do {                                                                      \
+    if (buf) { ROC_SPMME_L(TEAM_, true); }                                  \
+    else { ROC_SPMME_L(TEAM_, false); }                                     \
+  } while (0)
+  switch (team) {
+    case 8:  ROC_SPMME_T(8);  break;
+    case 16: ROC_SPMME_T(16); break;
+    case 32: ROC_SPMME_T(32); break;
+    default: ROC_SPMME_T(64);
+  }
+#undef ROC_SPMME_T
+#undef ROC_SPMME_L
+}
+
+template <typename T>
+void launch_edge_dot(float* dw, const T* dy, const T* x,
+                     const int64_t* rowptr, const int* colidx, int num_rows,
+                     int64_t D, hipStream_t stream) {
+  constexpr int EPU = EltTraits<T>::kPerVec;
+  const int64_t units = (D + EPU - 1) / EPU;
+  int team = 8;
+  while (team < units && team < 64) team *= 2;
+  const int tpb = kBlock / team;
+  dim3 grid(roc_grid_1d(num_rows, tpb, 8192));
+#define ROC_EDOT_L(TEAM_)                                                   \
+  hipLaunchKernelGGL((edge_dot_kernel<T, TEAM_>), grid, dim3(kBlock), 0,    \
+                     stream, dw, dy, x, rowptr, colidx, num_rows, D)
+  switch (team) {
+    case 8:  ROC_EDOT_L(8);  break;
+    case 16: ROC_EDOT_L(16); break;
+    case 32: ROC_EDOT_L(32); break;
+    default: ROC_EDOT_L(64);
+  }
+#undef ROC_EDOT_L
+}
+
 }  // namespace
 
 // A/B harness hook (scripts/bench_spmm.py): re-read the geometry env
@@ -392,6 +590,79 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
                        stream);
   } else {
     TORCH_CHECK(false, "spmm: unsupported dtype (bf16/f32 only)");
+  }
+  ROC_HIP_CHECK(hipGetLastError());
+}
+
+void spmm_edge(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
+               torch::Tensor colidx, torch::Tensor edge_val,
+               c10::optional<torch::Tensor> deg_dst,
+               c10::optional<torch::Tensor> row_order, bool accumulate) {
+  ROC_CHECK_DEV_CONT(out);
+  ROC_CHECK_DEV_CONT(x);
+  ROC_CHECK_DEV_CONT(rowptr);
+  ROC_CHECK_DEV_CONT(colidx);
+  ROC_CHECK_DEV_CONT(edge_val);
+  TORCH_CHECK(rowptr.scalar_type() == torch::kInt64, "rowptr must be int64");
+  TORCH_CHECK(colidx.scalar_type() == torch::kInt32, "colidx must be int32");
+  TORCH_CHECK(edge_val.scalar_type() == torch::kFloat32,
+              "edge_val must be fp32");
+  TORCH_CHECK(edge_val.numel() == colidx.numel(), "edge_val size mismatch");
+  TORCH_CHECK(out.scalar_type() == x.scalar_type(), "dtype mismatch");
+  const int num_rows = (int)out.size(0);
+  const int64_t D = out.size(1);
+  TORCH_CHECK(x.size(1) == D, "feature dim mismatch");
+  TORCH_CHECK(rowptr.size(0) == num_rows + 1, "rowptr size mismatch");
+  const float* dd =
+      deg_dst.has_value() ? deg_dst->data_ptr<float>() : nullptr;
+  const int* ro =
+      row_order.has_value() ? row_order->data_ptr<int>() : nullptr;
+  auto stream = roc_stream();
+  if (x.scalar_type() == torch::kBFloat16) {
+    launch_spmm_edge<unsigned short>(
+        (unsigned short*)out.data_ptr(), (const unsigned short*)x.data_ptr(),
+        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(),
+        edge_val.data_ptr<float>(), dd, ro, num_rows, D, accumulate,
+        (size_t)x.numel(), stream);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    launch_spmm_edge<float>(
+        out.data_ptr<float>(), x.data_ptr<float>(),
+        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(),
+        edge_val.data_ptr<float>(), dd, ro, num_rows, D, accumulate,
+        (size_t)x.numel(), stream);
+  } else {
+    TORCH_CHECK(false, "spmm_edge: unsupported dtype (bf16/f32 only)");
+  }
+  ROC_HIP_CHECK(hipGetLastError());
+}
+
+void edge_dot(torch::Tensor dw, torch::Tensor dy, torch::Tensor x,
+              torch::Tensor rowptr, torch::Tensor colidx) {
+  ROC_CHECK_DEV_CONT(dw);
+  ROC_CHECK_DEV_CONT(dy);
+  ROC_CHECK_DEV_CONT(x);
+  ROC_CHECK_DEV_CONT(rowptr);
+  ROC_CHECK_DEV_CONT(colidx);
+  TORCH_CHECK(dw.scalar_type() == torch::kFloat32, "dw must be fp32");
+  TORCH_CHECK(dw.numel() == colidx.numel(), "dw size mismatch");
+  TORCH_CHECK(dy.scalar_type() == x.scalar_type(), "dtype mismatch");
+  const int num_rows = (int)dy.size(0);
+  const int64_t D = dy.size(1);
+  TORCH_CHECK(x.size(1) == D, "feature dim mismatch");
+  TORCH_CHECK(rowptr.size(0) == num_rows + 1, "rowptr size mismatch");
+  auto stream = roc_stream();
+  if (dy.scalar_type() == torch::kBFloat16) {
+    launch_edge_dot<unsigned short>(
+        dw.data_ptr<float>(), (const unsigned short*)dy.data_ptr(),
+        (const unsigned short*)x.data_ptr(), rowptr.data_ptr<int64_t>(),
+        colidx.data_ptr<int>(), num_rows, D, stream);
+  } else if (dy.scalar_type() == torch::kFloat32) {
+    launch_edge_dot<float>(
+        dw.data_ptr<float>(), dy.data_ptr<float>(), x.data_ptr<float>(),
+        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), num_rows, D,
+        stream);
+  } else {
+    TORCH_CHECK(false, "edge_dot: unsupported dtype (bf16/f32 only)");
   }
   ROC_HIP_CHECK(hipGetLastError());
 }

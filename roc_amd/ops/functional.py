@@ -89,6 +89,78 @@ class _SpMM(torch.autograd.Function):
         return (dx,) + (None,) * 10
 
 
+class _SpMMEdge(torch.autograd.Function):
+    """out[v] = sum_{e in row v} w[e] * x[col_e]  (+ optional dst row
+    scale). The edge-tensor consumer op: w is a per-local-edge tensor
+    in CSR edge order (the reference declares edge tensors,
+    `gnn.cc:475-623`, but ships no op over them — this makes the
+    surface real). Backward: dx rides the transpose CSR with permuted
+    weights; dw[e] = <dy[row_e], x[col_e]>."""
+
+    @staticmethod
+    def forward(ctx, x, w, rowptr, colidx, t_rowptr, t_colidx, t_eperm,
+                num_rows, num_ext, dst_scale, row_order, t_row_order):
+        w = w.contiguous()
+        ctx.save_for_backward(x, w, rowptr, colidx, t_rowptr, t_colidx,
+                              t_eperm, dst_scale, t_row_order)
+        ctx.num_ext = num_ext
+        if _hip(x):
+            out = torch.empty(num_rows, x.shape[1], dtype=x.dtype,
+                              device=x.device)
+            _C.spmm_edge(out, x, rowptr, colidx, w, dst_scale, row_order,
+                         False)
+        else:
+            out = ref.spmm_weighted(x, rowptr, colidx, w, num_rows)
+            if dst_scale is not None:
+                out = out * dst_scale.unsqueeze(1).to(out.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x, w, rowptr, colidx, t_rowptr, t_colidx, t_eperm, dst_scale,
+         t_row_order) = ctx.saved_tensors
+        dy = dy.contiguous()
+        need_x, need_w = ctx.needs_input_grad[0], ctx.needs_input_grad[1]
+        dx = dw = None
+        if _hip(dy):
+            if dst_scale is not None:
+                tmp = torch.empty_like(dy)
+                _C.rowscale(tmp, dy, dst_scale)
+                dy = tmp
+            if need_x:
+                t_w = w[t_eperm].contiguous()
+                dx = torch.empty(ctx.num_ext, dy.shape[1], dtype=dy.dtype,
+                                 device=dy.device)
+                _C.spmm_edge(dx, dy, t_rowptr, t_colidx, t_w, None,
+                             t_row_order, False)
+            if need_w:
+                dw = torch.empty_like(w)
+                _C.edge_dot(dw, dy, x, rowptr, colidx)
+        else:
+            if dst_scale is not None:
+                dy = dy * dst_scale.unsqueeze(1).to(dy.dtype)
+            if need_x:
+                dx = ref.spmm_weighted(dy, t_rowptr, t_colidx, w[t_eperm],
+                                       ctx.num_ext)
+            if need_w:
+                dw = ref.edge_dot(dy, x, rowptr, colidx).to(w.dtype)
+        return (dx, dw) + (None,) * 10
+
+
+def scatter_gather_weighted(x, w, shard, dst_scale=None):
+    """Per-edge-weighted neighbor aggregation over the shard's local CSR.
+
+    x: [n_ext, D] halo-extended features; w: fp32 [num_local_edges] in
+    CSR edge order (see roc_amd.edge_tensor). Returns [n_local, D].
+    Both x and w receive gradients. Works at any world size — edge
+    tensors are partition-local, so no extra communication beyond the
+    usual halo exchange of x."""
+    return _SpMMEdge.apply(
+        x, w, shard.rowptr, shard.colidx, shard.t_rowptr, shard.t_colidx,
+        shard.t_edge_perm(), shard.n_local, shard.n_ext, dst_scale,
+        shard.row_order, shard.t_row_order)
+
+
 def scatter_gather(x, shard, normalize: bool = False, dst_scale=None,
                    src_scale=None):
     """Neighbor sum-aggregation over the shard's local CSR.

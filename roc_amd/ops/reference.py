@@ -27,6 +27,34 @@ def spmm(x: torch.Tensor, rowptr: torch.Tensor, colidx: torch.Tensor,
     return out
 
 
+def spmm_weighted(x: torch.Tensor, rowptr: torch.Tensor,
+                  colidx: torch.Tensor, edge_val: torch.Tensor,
+                  num_rows: int) -> torch.Tensor:
+    """out[v] = sum_{e in row v} edge_val[e] * x[col_e]  — the edge-tensor
+    consumer op (the reference declares per-edge tensors,
+    `gnn.cc:475-623` create_edge_tensor, but ships no op over them)."""
+    deg = (rowptr[1:] - rowptr[:-1]).to(torch.long)
+    dst = torch.repeat_interleave(
+        torch.arange(num_rows, dtype=torch.long, device=x.device), deg
+    )
+    out = torch.zeros(num_rows, x.shape[1], dtype=x.dtype, device=x.device)
+    out.index_add_(0, dst,
+                   x[colidx.to(torch.long)] * edge_val.unsqueeze(1).to(x.dtype))
+    return out
+
+
+def edge_dot(dy: torch.Tensor, x: torch.Tensor, rowptr: torch.Tensor,
+             colidx: torch.Tensor) -> torch.Tensor:
+    """dval[e] = <dy[row_e], x[col_e]> — gradient of spmm_weighted wrt
+    the per-edge values (fp32 accumulate)."""
+    num_rows = rowptr.numel() - 1
+    deg = (rowptr[1:] - rowptr[:-1]).to(torch.long)
+    dst = torch.repeat_interleave(
+        torch.arange(num_rows, dtype=torch.long, device=dy.device), deg
+    )
+    return (dy[dst].float() * x[colidx.to(torch.long)].float()).sum(dim=1)
+
+
 def degnorm(x: torch.Tensor, deg: torch.Tensor) -> torch.Tensor:
     """out[v] = x[v] / sqrt(indeg(v))  (reference `graphnorm_kernel.cu:19-57`)."""
     return x * torch.rsqrt(deg.clamp(min=1.0)).unsqueeze(1).to(x.dtype)

@@ -111,6 +111,56 @@ class GraphShard:
     def has_overlap_split(self) -> bool:
         return self.loc_rowptr is not None
 
+    @property
+    def num_local_edges(self) -> int:
+        """Size of this rank's edge window — edge tensors are
+        [num_local_edges, ...] in the local CSR's edge order
+        (reference edge-range partition, `gnn.cc:545-589`)."""
+        return int(self.colidx.numel())
+
+    def t_edge_perm(self) -> torch.Tensor:
+        """int64 [E_local]: for transposed edge j, its index in the
+        forward edge order (both the native counting-sort transpose and
+        the numpy/torch stable argsort produce the same stable order).
+        Lets edge values ride the backward SpMM: t_val = val[perm].
+        Lazily built and cached; the cache does not survive .to()."""
+        p = getattr(self, "_t_eperm", None)
+        if p is None:
+            p = torch.argsort(self.colidx.long(), stable=True)
+            object.__setattr__(self, "_t_eperm", p)
+        return p
+
+
+def edge_tensor(shard: GraphShard, dim: Optional[int] = None,
+                dtype=torch.float32, init: str = "zeros") -> torch.Tensor:
+    """Per-edge tensor for this rank's edge window, aligned with the
+    local CSR edge order (the analog of the reference's
+    `create_edge_tensor`, `gnn.cc:475-623`: EDGE-typed tensors
+    partitioned by the graph's edge ranges; its live driver never
+    consumed them — here `ops.functional.scatter_gather_weighted` does).
+
+    dim=None -> [E_local] (scalar per edge); else [E_local, dim].
+    init: "zeros" | "ones" | "gcn_norm" (1/sqrt(deg_dst*deg_src) per
+    edge — with these weights the weighted aggregation equals the fused
+    symmetric-norm GCN path; scalar only).
+    """
+    e = shard.num_local_edges
+    dev = shard.colidx.device
+    shape = (e,) if dim is None else (e, dim)
+    if init == "zeros":
+        return torch.zeros(*shape, dtype=dtype, device=dev)
+    if init == "ones":
+        return torch.ones(*shape, dtype=dtype, device=dev)
+    if init == "gcn_norm":
+        assert dim is None, "gcn_norm is a scalar-per-edge init"
+        deg = (shard.rowptr[1:] - shard.rowptr[:-1])
+        row_of_edge = torch.repeat_interleave(
+            torch.arange(shard.n_local, dtype=torch.long, device=dev), deg)
+        val = (shard.rsqrt_deg_local[row_of_edge]
+               * shard.rsqrt_deg_ext[shard.colidx.long()])
+        return val.to(dtype)
+    raise ValueError(f"unknown edge_tensor init {init!r}")
+
 
 def edge_balanced_bounds(rowptr: torch.Tensor, num_parts: int) -> List[int]:
     """Contiguous vertex ranges with ~equal in-edge counts
