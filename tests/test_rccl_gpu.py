@@ -38,7 +38,7 @@ def _device(rank) -> str:
     return f"cuda:{rank % torch.cuda.device_count()}"
 
 
-def _init(rank, port, comm_mode="halo"):
+def _init(rank, port, comm_mode="halo", ws=WS):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["ROC_COMM_MODE"] = comm_mode
@@ -48,7 +48,7 @@ def _init(rank, port, comm_mode="halo"):
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     torch.cuda.set_device(torch.device(_device(rank)))
     dist.init_process_group(
-        "nccl", rank=rank, world_size=WS,
+        "nccl", rank=rank, world_size=ws,
         timeout=datetime.timedelta(seconds=120))
 
 
@@ -88,8 +88,7 @@ def _run(worker, port, timeout=300):
 
 def _graph_capture_worker(rank, port, q):
     try:
-        _init(rank, port)
-        import torch.distributed as dist
+        _init(rank, port, ws=1)  # single-process RCCL group
         from roc_amd import build_model, AdamOptimizer, Trainer
         from roc_amd.graph import synthetic_dataset
         from roc_amd.parallel.partition import build_shard
