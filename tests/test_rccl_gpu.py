@@ -38,13 +38,16 @@ def _device(rank) -> str:
     return f"cuda:{rank % torch.cuda.device_count()}"
 
 
-def _init(rank, port, comm_mode="halo", ws=WS):
+def _init(rank, port, comm_mode="halo", ws=WS, blocking_wait=True):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["ROC_COMM_MODE"] = comm_mode
     # fail fast instead of hanging the leased box: collectives abort
-    # after the timeout when blocking-wait is on
-    os.environ["TORCH_NCCL_BLOCKING_WAIT"] = "1"
+    # after the timeout when blocking-wait is on. NEVER combine with
+    # hipGraph capture: a blocking-wait collective inside a capture
+    # spins forever (capture records, nothing executes — measured as a
+    # 400 s hang in r2c5).
+    os.environ["TORCH_NCCL_BLOCKING_WAIT"] = "1" if blocking_wait else "0"
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     torch.cuda.set_device(torch.device(_device(rank)))
     dist.init_process_group(
@@ -88,7 +91,9 @@ def _run(worker, port, timeout=300):
 
 def _graph_capture_worker(rank, port, q):
     try:
-        _init(rank, port, ws=1)  # single-process RCCL group
+        # single-process RCCL group; blocking-wait OFF (incompatible
+        # with capture — see _init docstring)
+        _init(rank, port, ws=1, blocking_wait=False)
         from roc_amd import build_model, AdamOptimizer, Trainer
         from roc_amd.graph import synthetic_dataset
         from roc_amd.parallel.partition import build_shard
@@ -160,16 +165,20 @@ def _graph_capture_worker(rank, port, q):
 
 def test_rccl_graph_capture_ws1():
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
     p = ctx.Process(target=_graph_capture_worker, args=(0, 29609, q))
     p.start()
-    p.join(timeout=300)
+    p.join(timeout=240)
     if p.is_alive():
         p.terminate()
         p.join(10)
         raise AssertionError("graph-capture worker hung (capture of RCCL "
                              "collective did not complete)")
-    rank, payload, err = q.get()
+    import queue as _queue
+    try:
+        rank, payload, err = q.get(timeout=10)
+    except _queue.Empty:
+        raise AssertionError("graph-capture worker died without reporting")
     assert err is None, err
     w_eager, w_graph, still_graph, raw_ok = payload
     # capture must have survived (no silent eager fallback) and the
