@@ -67,6 +67,18 @@ class Trainer:
     # Philox dropout offsets and the Adam bias-corrected/decayed step size.
     def enable_graph_capture(self, warmup_epochs: int = 2):
         assert self.device.type == "cuda", "graph capture needs a GPU"
+        import os
+        if self.shard.world_size > 1 and any(
+                os.environ.get(v) == "1"
+                for v in ("TORCH_NCCL_BLOCKING_WAIT", "NCCL_BLOCKING_WAIT")):
+            # a blocking-wait collective inside a capture spins forever
+            # (capture records, nothing executes — measured, r2c5);
+            # refuse capture instead of hanging the job
+            import sys
+            print("[roc_amd] NCCL blocking-wait is on; hipGraph capture "
+                  "of collectives would hang — staying eager",
+                  file=sys.stderr, flush=True)
+            return
         self._step_dev = torch.zeros(1, dtype=torch.int64, device=self.device)
         F.set_dropout_counter(self._step_dev)
         self.optimizer.set_device_step(self._step_dev)
