@@ -91,9 +91,12 @@ def _run(worker, port, timeout=300):
 
 def _graph_capture_worker(rank, port, q):
     try:
+        import faulthandler
+        faulthandler.enable()
         # single-process RCCL group; blocking-wait OFF (incompatible
         # with capture — see _init docstring)
         _init(rank, port, ws=1, blocking_wait=False)
+        q.put(("stage", "init", None))
         from roc_amd import build_model, AdamOptimizer, Trainer
         from roc_amd.graph import synthetic_dataset
         from roc_amd.parallel.partition import build_shard
@@ -124,7 +127,9 @@ def _graph_capture_worker(rank, port, q):
                     tr.use_graph)
 
         w_eager, _ = run(False)
+        q.put(("stage", "eager_done", None))
         w_graph, still_graph = run(True)
+        q.put(("stage", "capture_done", None))
 
         # raw capture of every collective type the multi-GPU epoch uses
         # (a2av halo exchange, all_gather fwd, reduce-to-owner bwd)
@@ -155,9 +160,9 @@ def _graph_capture_worker(rank, port, q):
         torch.cuda.synchronize()
         raw_ok = (float(x[0]) == 2.0 and float(y[0]) == 2.0
                   and float(gat[0]) == 2.0 and float(red[0]) == 2.0)
-        q.put((rank, (w_eager, w_graph, still_graph, raw_ok), None))
+        q.put(("result", (w_eager, w_graph, still_graph, raw_ok), None))
     except Exception as e:  # noqa: BLE001
-        q.put((rank, None, repr(e)))
+        q.put(("error", None, repr(e)))
     finally:
         if dist.is_initialized():
             dist.destroy_process_group()
@@ -169,18 +174,27 @@ def test_rccl_graph_capture_ws1():
     p = ctx.Process(target=_graph_capture_worker, args=(0, 29609, q))
     p.start()
     p.join(timeout=240)
-    if p.is_alive():
+    hung = p.is_alive()
+    if hung:
         p.terminate()
         p.join(10)
-        raise AssertionError("graph-capture worker hung (capture of RCCL "
-                             "collective did not complete)")
     import queue as _queue
+    msgs = []
     try:
-        rank, payload, err = q.get(timeout=10)
+        while True:
+            msgs.append(q.get(timeout=5))
     except _queue.Empty:
-        raise AssertionError("graph-capture worker died without reporting")
-    assert err is None, err
-    w_eager, w_graph, still_graph, raw_ok = payload
+        pass
+    stages = [m[1] for m in msgs if m[0] == "stage"]
+    if hung:
+        raise AssertionError(
+            f"graph-capture worker hung after stages {stages}")
+    errs = [m[2] for m in msgs if m[0] == "error"]
+    assert not errs, errs
+    results = [m[1] for m in msgs if m[0] == "result"]
+    assert results, (f"worker died without reporting (exitcode "
+                     f"{p.exitcode}, stages {stages})")
+    w_eager, w_graph, still_graph, raw_ok = results[0]
     # capture must have survived (no silent eager fallback) and the
     # replayed epochs must match the eager run bit-for-bit in fp32
     assert still_graph, "hipGraph capture fell back to eager"
