@@ -139,9 +139,13 @@ def main():
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
                       compute_dtype=dtype, grad_scale=1.0, seed=args.seed,
                       num_classes=num_classes, local_slices=True)
-    # hipGraph capture is default-on for 1 GPU; for multi-GPU it wraps
-    # RCCL collectives in the graph — enable explicitly once measured
-    # (ROC_GRAPH_MULTI=1). Capture failure falls back to eager either way.
+    # hipGraph capture is default-on for 1 GPU. For multi-GPU it stays
+    # OFF by MEASUREMENT (r2c7): RCCL a2av/all_gather/reduce inside a
+    # captured graph segfault on this stack (only all_reduce capture
+    # works), and the eager epoch costs just 0.44 ms of Python/launch
+    # time (eager 20.38 ms vs captured 20.15 ms at ws=1) — so eager
+    # collectives are safe AND cheap. ROC_GRAPH_MULTI=1 forces capture
+    # for future stacks where RCCL-in-graph works.
     if device != "cpu" and not args.no_graph and (
             world == 1 or os.environ.get("ROC_GRAPH_MULTI") == "1"):
         trainer.enable_graph_capture()

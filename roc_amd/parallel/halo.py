@@ -79,7 +79,15 @@ import os as _os
 
 
 def overlap_enabled() -> bool:
-    return _os.environ.get("ROC_OVERLAP", "0") == "1"
+    """Halo-mode comm/compute overlap (interior SpMM while the a2av is
+    in flight). Default ON since round 2: correctness is equality-
+    tested vs the sequential path at world sizes 2/3/8 (gloo), the
+    collectives run eagerly (no hipGraph interaction — RCCL a2av
+    inside a captured graph segfaults on this stack, see
+    tests/test_rccl_gpu.py::test_rccl_raw_collectives_in_graph), and
+    hiding the exchange behind the interior aggregation is the whole
+    point of the split. ROC_OVERLAP=0 reverts."""
+    return _os.environ.get("ROC_OVERLAP", "1") == "1"
 
 
 def _spmm_part(out, x, rowptr, colidx, dst, acc, order=None):
