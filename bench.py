@@ -61,9 +61,6 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     on_gpu = torch.cuda.is_available()
     if world > 1:
-        # nccl IS RCCL on ROCm; gloo keeps the path testable on CPU boxes
-        dist.init_process_group("nccl" if on_gpu else "gloo",
-                                rank=rank, world_size=world)
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         # ROC_DEVICE_OVERRIDE: testing hook. NOTE: RCCL refuses two
         # ranks on ONE device (measured: "Duplicate GPU detected",
@@ -73,10 +70,16 @@ def main():
         dev_idx = int(os.environ.get("ROC_DEVICE_OVERRIDE", local_rank))
         if on_gpu:
             dev_idx %= max(torch.cuda.device_count(), 1)
+            # pin BEFORE init so the RCCL communicator binds the right
+            # device (no rank->device guessing)
             torch.cuda.set_device(dev_idx)
             device = f"cuda:{dev_idx}"
         else:
             device = "cpu"
+        # nccl IS RCCL on ROCm; gloo keeps the path testable on CPU boxes
+        dist.init_process_group(
+            "nccl" if on_gpu else "gloo", rank=rank, world_size=world,
+            device_id=torch.device(device) if on_gpu else None)
     else:
         device = "cuda:0" if on_gpu else "cpu"
         if device != "cpu":

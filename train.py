@@ -149,12 +149,14 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     on_gpu = torch.cuda.is_available()
     if world > 1:
-        dist.init_process_group("nccl" if on_gpu else "gloo",
-                                rank=rank, world_size=world)
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         if on_gpu:
+            # pin BEFORE init so RCCL binds the right device
             torch.cuda.set_device(local_rank)
         device = f"cuda:{local_rank}" if on_gpu else "cpu"
+        dist.init_process_group(
+            "nccl" if on_gpu else "gloo", rank=rank, world_size=world,
+            device_id=torch.device(device) if on_gpu else None)
     else:
         device = "cuda:0" if on_gpu else "cpu"
 
