@@ -59,6 +59,34 @@ def _single_rank_baseline():
     return tr.evaluate(), model.weights[0].detach()
 
 
+def test_bench_contract_ws8_gloo():
+    """The driver's largest scale point: torchrun ws=8 through bench.py
+    end-to-end (gloo on CPU, tiny scale) — one JSON line, whole-job
+    metric, agreed auto-extended step count across 8 ranks."""
+    import json
+    import subprocess
+    import sys
+    import roc_amd
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(
+        roc_amd.__file__)))
+    env = dict(os.environ)
+    env.pop("ROC_SPMM_SCHEDULE", None)
+    env["ROC_BENCH_CACHE"] = env.get("TMPDIR", "/tmp")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29739", os.path.join(repo, "bench.py"),
+         "--gpus", "8", "--steps", "2", "--warmup", "1", "--scale",
+         "0.02", "--exact-steps"],
+        capture_output=True, text=True, timeout=900, env=env, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 8 and out["steps"] == 2
+    assert out["unit"] == "s/epoch" and out["value"] > 0
+    assert "x8" in out["config"]["parallelism"]
+
+
 @pytest.mark.parametrize("mode", ["halo", "allgather"])
 def test_ws8_matches_single_rank(mode):
     ctx = mp.get_context("spawn")
