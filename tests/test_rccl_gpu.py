@@ -191,7 +191,12 @@ def test_rccl_allreduce_in_graph_ws1():
                         f"died (exitcode {p.exitcode})"))
     diff, still_graph = results[0]
     assert still_graph, "hipGraph capture fell back to eager"
-    assert diff < 1e-6, diff
+    # graph mode computes the Adam bias-corrected step size ON DEVICE
+    # (powf) while eager computes it on host (libm) — a ~1-ulp alpha_t
+    # difference that compounds to ~4e-6 over 8 fp32 epochs (measured
+    # r2c9). Anything larger means the captured collective or replay
+    # schedule is actually wrong.
+    assert diff < 5e-5, diff
 
 
 def _raw_capture_worker(rank, port, which, q):
