@@ -31,6 +31,10 @@ void spmm_edge(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
                c10::optional<torch::Tensor> row_order, bool accumulate);
 void edge_dot(torch::Tensor dw, torch::Tensor dy, torch::Tensor x,
               torch::Tensor rowptr, torch::Tensor colidx);
+void edge_softmax_fwd(torch::Tensor alpha, torch::Tensor s,
+                      torch::Tensor rowptr);
+void edge_softmax_bwd(torch::Tensor ds, torch::Tensor dalpha,
+                      torch::Tensor alpha, torch::Tensor rowptr);
 void register_graph_cpu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -75,5 +79,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("accumulate") = false);
   m.def("edge_dot", &edge_dot,
         "dw[e] = <dy[row_e], x[col_e]> (edge-value gradient)");
+  m.def("edge_softmax_fwd", &edge_softmax_fwd,
+        "per-row segment softmax over edge scores (GAT attention)");
+  m.def("edge_softmax_bwd", &edge_softmax_bwd);
   register_graph_cpu(m);
 }

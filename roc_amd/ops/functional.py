@@ -161,6 +161,52 @@ def scatter_gather_weighted(x, w, shard, dst_scale=None):
         shard.row_order, shard.t_row_order)
 
 
+class _EdgeSoftmax(torch.autograd.Function):
+    """Per-destination-row segment softmax over edge scores (the GAT
+    attention normalizer). GPU: one fused CDNA4 kernel per direction
+    (edge_softmax.hip); CPU: index-op reference."""
+
+    @staticmethod
+    def forward(ctx, s, rowptr, row_of_edge):
+        s = s.contiguous()
+        if _hip(s):
+            alpha = torch.empty_like(s)
+            _C.edge_softmax_fwd(alpha, s, rowptr)
+        else:
+            n = rowptr.numel() - 1
+            m = torch.full((n,), float("-inf"), dtype=s.dtype)
+            m = m.scatter_reduce(0, row_of_edge, s, reduce="amax",
+                                 include_self=True)
+            ex = (s - m[row_of_edge]).exp()
+            denom = torch.zeros(n, dtype=s.dtype).index_add_(
+                0, row_of_edge, ex)
+            alpha = ex / denom[row_of_edge]
+        ctx.save_for_backward(alpha, rowptr, row_of_edge)
+        return alpha
+
+    @staticmethod
+    def backward(ctx, dalpha):
+        alpha, rowptr, row_of_edge = ctx.saved_tensors
+        dalpha = dalpha.contiguous()
+        if _hip(dalpha):
+            ds = torch.empty_like(dalpha)
+            _C.edge_softmax_bwd(ds, dalpha, alpha, rowptr)
+        else:
+            n = rowptr.numel() - 1
+            dot = torch.zeros(n, dtype=alpha.dtype).index_add_(
+                0, row_of_edge, alpha * dalpha)
+            ds = alpha * (dalpha - dot[row_of_edge])
+        return ds, None, None
+
+
+def edge_softmax(scores, shard):
+    """softmax of fp32 edge scores within each destination row's
+    in-edge segment: Σ_{e∈row v} out[e] = 1. Differentiable; pairs
+    with scatter_gather_weighted for attention-style aggregation."""
+    return _EdgeSoftmax.apply(scores.float(), shard.rowptr,
+                              shard.row_of_edge())
+
+
 def scatter_gather(x, shard, normalize: bool = False, dst_scale=None,
                    src_scale=None):
     """Neighbor sum-aggregation over the shard's local CSR.

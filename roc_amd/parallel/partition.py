@@ -118,6 +118,18 @@ class GraphShard:
         (reference edge-range partition, `gnn.cc:545-589`)."""
         return int(self.colidx.numel())
 
+    def row_of_edge(self) -> torch.Tensor:
+        """int64 [E_local]: destination row of each local edge (lazily
+        cached; does not survive .to())."""
+        r = getattr(self, "_row_of_edge", None)
+        if r is None:
+            deg = (self.rowptr[1:] - self.rowptr[:-1]).long()
+            r = torch.repeat_interleave(
+                torch.arange(self.n_local, dtype=torch.long,
+                             device=self.rowptr.device), deg)
+            object.__setattr__(self, "_row_of_edge", r)
+        return r
+
     def t_edge_perm(self) -> torch.Tensor:
         """int64 [E_local]: for transposed edge j, its index in the
         forward edge order (both the native counting-sort transpose and
