@@ -20,4 +20,4 @@ from .parallel.partition import (GraphShard, build_shard,  # noqa: F401
                                  edge_balanced_bounds, edge_tensor)
 from .optim import AdamOptimizer  # noqa: F401
 from .engine import Trainer  # noqa: F401
-from .models import build_model, GCN, GraphSAGE, GIN  # noqa: F401
+from .models import build_model, GCN, GraphSAGE, GIN, GAT  # noqa: F401

@@ -1,6 +1,7 @@
 from .gcn import GCN
 from .sage import GraphSAGE
 from .gin import GIN
+from .gat import GAT
 
 
 def build_model(name: str, dims, dropout: float = 0.5, seed: int = 1, **kw):
@@ -11,4 +12,6 @@ def build_model(name: str, dims, dropout: float = 0.5, seed: int = 1, **kw):
         return GraphSAGE(dims, dropout=dropout, seed=seed, **kw)
     if name == "gin":
         return GIN(dims, dropout=dropout, seed=seed, **kw)
+    if name == "gat":
+        return GAT(dims, dropout=dropout, seed=seed, **kw)
     raise ValueError(f"unknown model {name!r}")

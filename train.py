@@ -10,7 +10,7 @@ Multi GPU (one process per GPU, RCCL):
 Reference flag parity: --file (.lux dataset prefix), --layers D0-D1-...-C,
 --epochs, --lr, --weight-decay, --decay-rate, --decay-steps, --dropout,
 --seed; metrics printed every 5 epochs (`gnn.cc:107-110`). New:
---model gcn|sage|gin, --dtype, --checkpoint/--resume, --trace.
+--model gcn|sage|gin|gat, --dtype, --checkpoint/--resume, --trace.
 """
 import argparse
 import os
@@ -45,7 +45,7 @@ def parse_args():
                          "e.g. 602-256-41")
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--num-layers", type=int, default=2)
-    ap.add_argument("--model", default="gcn", choices=["gcn", "sage", "gin"])
+    ap.add_argument("--model", default="gcn", choices=["gcn", "sage", "gin", "gat"])
     ap.add_argument("--epochs", "-e", type=int, default=100)
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--weight-decay", "--wd", type=float, default=1e-4)
