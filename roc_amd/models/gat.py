@@ -71,7 +71,7 @@ class GAT(torch.nn.Module):
         h = F.dropout(x, self.p, self.training, call_id=i)
         h = F.linear(h, self.weights[i])            # [n_local, nh*dh]
         h_ext = halo_exchange(h, shard, group)      # [n_ext, nh*dh]
-        col = shard.colidx.long()
+        col = shard.colidx_long()
         row = shard.row_of_edge()
         outs = []
         for k in range(nh):

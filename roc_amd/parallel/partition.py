@@ -118,6 +118,15 @@ class GraphShard:
         (reference edge-range partition, `gnn.cc:545-589`)."""
         return int(self.colidx.numel())
 
+    def colidx_long(self) -> torch.Tensor:
+        """int64 view of colidx (lazily cached — torch index ops need
+        int64 and a fresh .long() would allocate E*8 B per call)."""
+        c = getattr(self, "_colidx_long", None)
+        if c is None:
+            c = self.colidx.long()
+            object.__setattr__(self, "_colidx_long", c)
+        return c
+
     def row_of_edge(self) -> torch.Tensor:
         """int64 [E_local]: destination row of each local edge (lazily
         cached; does not survive .to())."""
