@@ -35,6 +35,14 @@ void edge_softmax_fwd(torch::Tensor alpha, torch::Tensor s,
                       torch::Tensor rowptr);
 void edge_softmax_bwd(torch::Tensor ds, torch::Tensor dalpha,
                       torch::Tensor alpha, torch::Tensor rowptr);
+void att_softmax_fwd(torch::Tensor alpha, torch::Tensor s_src,
+                     torch::Tensor s_dst, torch::Tensor rowptr,
+                     torch::Tensor colidx, double slope);
+void att_softmax_bwd(torch::Tensor dsrc, torch::Tensor dsdst,
+                     torch::Tensor dalpha, torch::Tensor alpha,
+                     torch::Tensor s_src, torch::Tensor s_dst,
+                     torch::Tensor rowptr, torch::Tensor colidx,
+                     double slope);
 void register_graph_cpu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -82,5 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("edge_softmax_fwd", &edge_softmax_fwd,
         "per-row segment softmax over edge scores (GAT attention)");
   m.def("edge_softmax_bwd", &edge_softmax_bwd);
+  m.def("att_softmax_fwd", &att_softmax_fwd,
+        "fused GAT attention: lrelu(s_src[col]+s_dst[row]) -> softmax");
+  m.def("att_softmax_bwd", &att_softmax_bwd);
   register_graph_cpu(m);
 }

@@ -71,8 +71,6 @@ class GAT(torch.nn.Module):
         h = F.dropout(x, self.p, self.training, call_id=i)
         h = F.linear(h, self.weights[i])            # [n_local, nh*dh]
         h_ext = halo_exchange(h, shard, group)      # [n_ext, nh*dh]
-        col = shard.colidx_long()
-        row = shard.row_of_edge()
         outs = []
         for k in range(nh):
             hk = h_ext[:, k * dh:(k + 1) * dh]
@@ -80,9 +78,9 @@ class GAT(torch.nn.Module):
             s_src = (hk.float() @ self.a_src[i][k].float())   # [n_ext]
             s_dst = (hk[:shard.n_local].float()
                      @ self.a_dst[i][k].float())              # [n_local]
-            scores = torch.nn.functional.leaky_relu(
-                s_src[col] + s_dst[row], self.slope)
-            alpha = F.edge_softmax(scores, shard)
+            # fused: lrelu(s_src[col]+s_dst[row]) -> segment softmax,
+            # one kernel each direction (no E-length intermediates)
+            alpha = F.attention_softmax(s_src, s_dst, shard, self.slope)
             outs.append(F.scatter_gather_weighted(hk.contiguous(), alpha,
                                                   shard))
         out = outs[0] if nh == 1 else torch.cat(outs, dim=1)

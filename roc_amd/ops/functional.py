@@ -207,6 +207,70 @@ def edge_softmax(scores, shard):
                               shard.row_of_edge())
 
 
+class _AttentionSoftmax(torch.autograd.Function):
+    """alpha[e] = softmax_row(lrelu(s_src[col_e] + s_dst[row])) — the
+    whole GAT score+normalize chain in ONE kernel per direction
+    (edge_softmax.hip att_*). Replaces 4 E-length torch intermediates
+    and a 115M-element index_add backward. CPU path composes index
+    ops (the numerics oracle)."""
+
+    @staticmethod
+    def forward(ctx, s_src, s_dst, rowptr, colidx, row_of_edge, slope):
+        s_src = s_src.contiguous()
+        s_dst = s_dst.contiguous()
+        if _hip(s_src):
+            alpha = torch.empty(colidx.numel(), dtype=torch.float32,
+                                device=s_src.device)
+            _C.att_softmax_fwd(alpha, s_src, s_dst, rowptr, colidx, slope)
+        else:
+            n = rowptr.numel() - 1
+            sc = s_src[colidx.long()] + s_dst[row_of_edge]
+            sc = torch.nn.functional.leaky_relu(sc, slope)
+            m = torch.full((n,), float("-inf"), dtype=sc.dtype)
+            m = m.scatter_reduce(0, row_of_edge, sc, reduce="amax",
+                                 include_self=True)
+            ex = (sc - m[row_of_edge]).exp()
+            den = torch.zeros(n, dtype=sc.dtype).index_add_(
+                0, row_of_edge, ex)
+            alpha = ex / den[row_of_edge]
+        ctx.save_for_backward(alpha, s_src, s_dst, rowptr, colidx,
+                              row_of_edge)
+        ctx.slope = slope
+        return alpha
+
+    @staticmethod
+    def backward(ctx, dalpha):
+        alpha, s_src, s_dst, rowptr, colidx, row_of_edge = ctx.saved_tensors
+        slope = ctx.slope
+        dalpha = dalpha.contiguous()
+        if _hip(dalpha):
+            dsrc = torch.zeros_like(s_src)
+            dsdst = torch.empty_like(s_dst)
+            _C.att_softmax_bwd(dsrc, dsdst, dalpha, alpha, s_src, s_dst,
+                               rowptr, colidx, slope)
+        else:
+            n = rowptr.numel() - 1
+            dot = torch.zeros(n, dtype=alpha.dtype).index_add_(
+                0, row_of_edge, alpha * dalpha)
+            ds = alpha * (dalpha - dot[row_of_edge])
+            raw = s_src[colidx.long()] + s_dst[row_of_edge]
+            dscore = torch.where(raw > 0, ds, slope * ds)
+            dsrc = torch.zeros_like(s_src).index_add_(
+                0, colidx.long(), dscore)
+            dsdst = torch.zeros_like(s_dst).index_add_(
+                0, row_of_edge, dscore)
+        return dsrc, dsdst, None, None, None, None
+
+
+def attention_softmax(s_src, s_dst, shard, slope: float = 0.2):
+    """Fused GAT attention coefficients from per-node score halves:
+    s_src fp32 [n_ext] (source half, halo-extended), s_dst fp32
+    [n_local]. Returns alpha fp32 [E_local]; both inputs get grads."""
+    return _AttentionSoftmax.apply(s_src.float(), s_dst.float(),
+                                   shard.rowptr, shard.colidx,
+                                   shard.row_of_edge(), slope)
+
+
 def scatter_gather(x, shard, normalize: bool = False, dst_scale=None,
                    src_scale=None):
     """Neighbor sum-aggregation over the shard's local CSR.

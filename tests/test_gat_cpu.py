@@ -54,6 +54,30 @@ def test_edge_softmax_grad_matches_autograd():
         (s1.grad - s2.grad).abs().max()
 
 
+def test_attention_softmax_matches_composed():
+    """Fused score+lrelu+softmax vs the explicit torch composition,
+    forward AND both input grads."""
+    torch.manual_seed(7)
+    g, *_, sh = _setup()
+    col, row = sh.colidx_long(), sh.row_of_edge()
+    slope = 0.2
+    src1 = torch.randn(sh.n_ext if sh.n_halo else sh.n_local,
+                       requires_grad=True)
+    dst1 = torch.randn(sh.n_local, requires_grad=True)
+    src2 = src1.detach().clone().requires_grad_(True)
+    dst2 = dst1.detach().clone().requires_grad_(True)
+    gy = torch.randn(sh.num_local_edges)
+    a1 = F.attention_softmax(src1, dst1, sh, slope)
+    a1.backward(gy)
+    sc = torch.nn.functional.leaky_relu(src2[col] + dst2[row], slope)
+    a2 = _composed_softmax(sc, sh.rowptr, row)
+    a2.backward(gy)
+    assert torch.allclose(a1, a2, atol=1e-6)
+    assert torch.allclose(src1.grad, src2.grad, atol=1e-5), \
+        (src1.grad - src2.grad).abs().max()
+    assert torch.allclose(dst1.grad, dst2.grad, atol=1e-5)
+
+
 def test_gat_trains_and_attention_is_learned():
     torch.manual_seed(0)
     g, feats, labels, mask, c, sh = _setup()
