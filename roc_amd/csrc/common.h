@@ -79,6 +79,10 @@ __device__ __forceinline__ void store_bf16x8(unsigned short* p,
   v.w = pack_bf16(a[6], a[7]);
   *reinterpret_cast<uint4*>(p) = v;
 }
+__device__ __forceinline__ void store_f32x8(float* p, const float* a) {
+  store_f32x4(p, a);
+  store_f32x4(p + 4, a + 4);
+}
 
 __device__ __forceinline__ float elt_to_f32(float x) { return x; }
 __device__ __forceinline__ float elt_to_f32(unsigned short x) {

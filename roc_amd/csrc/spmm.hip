@@ -199,9 +199,12 @@ __device__ void row_accum_tail(
 // (a forced 6-waves/SIMD __launch_bounds__ hint was measured NEUTRAL:
 //  the allocator spills 28 B/lane on half the hot variants and any
 //  occupancy gain washes out — keep the default allocation)
-template <typename T, int TEAM, int UN, bool BUF, bool SRC>
+// OT (output type) may differ from T: OT=float with T=bf16 is the
+// strip-blocked accumulation path — fp32 partials in HBM, one rounding
+// at the final cast instead of one per strip pass.
+template <typename T, typename OT, int TEAM, int UN, bool BUF, bool SRC>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
-    T* __restrict__ out, const T* __restrict__ x,
+    OT* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
     const float* __restrict__ deg_dst, const float* __restrict__ deg_src,
     const int* __restrict__ row_order, int num_rows, int64_t D,
@@ -222,8 +225,8 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
       const int64_t e0 = rowptr[row];
       const int64_t e1 = rowptr[row + 1];
       float acc[EPU];
-      T* o = out + (int64_t)row * D + col0;
-      if (accumulate) {  // halo-overlap pass 2: start from the partials
+      OT* o = out + (int64_t)row * D + col0;
+      if (accumulate) {  // strip / halo-overlap pass 2: from partials
 #pragma unroll
         for (int j = 0; j < EPU; ++j) acc[j] = elt_to_f32(o[j]);
       } else {
@@ -245,13 +248,19 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
 #pragma unroll
         for (int j = 0; j < EPU; ++j) acc[j] *= s;
       }
-      if constexpr (EPU == 8) store_bf16x8(o, acc); else store_f32x4(o, acc);
+      if constexpr (std::is_same<OT, unsigned short>::value) {
+        store_bf16x8(o, acc);
+      } else if constexpr (EPU == 8) {
+        store_f32x8(o, acc);
+      } else {
+        store_f32x4(o, acc);
+      }
     }
   } else if (nvalid > 0) {
     for (int ri = team; ri < num_rows; ri += nteams) {
       const int row = row_order ? row_order[ri] : ri;
       float acc[EPU];
-      T* o = out + (int64_t)row * D + col0;
+      OT* o = out + (int64_t)row * D + col0;
       for (int j = 0; j < nvalid; ++j)
         acc[j] = accumulate ? elt_to_f32(o[j]) : 0.f;
       row_accum_tail<T, EPU>(acc, x, D, col0, nvalid, colidx, deg_src,
@@ -294,11 +303,12 @@ static const SpmmKnobs& spmm_knobs() {
   return g_spmm_knobs;
 }
 
-template <typename T>
-void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
-                 const float* deg_dst, const float* deg_src,
-                 const int* row_order, int num_rows, int64_t D,
-                 bool accumulate, size_t x_elems, hipStream_t stream) {
+template <typename T, typename OT>
+void launch_spmm(OT* out, const T* x, const int64_t* rowptr,
+                 const int* colidx, const float* deg_dst,
+                 const float* deg_src, const int* row_order, int num_rows,
+                 int64_t D, bool accumulate, size_t x_elems,
+                 hipStream_t stream) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const SpmmKnobs& kn = spmm_knobs();
   const int64_t units = (D + EPU - 1) / EPU;
@@ -319,12 +329,14 @@ void launch_spmm(T* out, const T* x, const int64_t* rowptr, const int* colidx,
 #define ROC_SPMM_L3(TEAM_, UN_, BUF_)                                       \
   do {                                                                      \
     if (deg_src) {                                                          \
-      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN_, BUF_, true>), grid,    \
+      hipLaunchKernelGGL((spmm_kernel<T, OT, TEAM_, UN_, BUF_, true>),     \
+                         grid,                                              \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
                          deg_dst, deg_src, row_order, num_rows, D,          \
                          accumulate, x_bytes);                              \
     } else {                                                                \
-      hipLaunchKernelGGL((spmm_kernel<T, TEAM_, UN_, BUF_, false>), grid,   \
+      hipLaunchKernelGGL((spmm_kernel<T, OT, TEAM_, UN_, BUF_, false>),    \
+                         grid,                                              \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
                          deg_dst, deg_src, row_order, num_rows, D,          \
                          accumulate, x_bytes);                              \
@@ -566,7 +578,10 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
   ROC_CHECK_DEV_CONT(colidx);
   TORCH_CHECK(rowptr.scalar_type() == torch::kInt64, "rowptr must be int64");
   TORCH_CHECK(colidx.scalar_type() == torch::kInt32, "colidx must be int32");
-  TORCH_CHECK(out.scalar_type() == x.scalar_type(), "dtype mismatch");
+  TORCH_CHECK(out.scalar_type() == x.scalar_type() ||
+                  (out.scalar_type() == torch::kFloat32 &&
+                   x.scalar_type() == torch::kBFloat16),
+              "dtype mismatch (out must match x, or be fp32 for bf16 x)");
   const int num_rows = (int)out.size(0);
   const int64_t D = out.size(1);
   TORCH_CHECK(x.size(1) == D, "feature dim mismatch");
@@ -578,13 +593,19 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
   const int* ro =
       row_order.has_value() ? row_order->data_ptr<int>() : nullptr;
   auto stream = roc_stream();
-  if (x.scalar_type() == torch::kBFloat16) {
-    launch_spmm<unsigned short>(
+  if (x.scalar_type() == torch::kBFloat16 &&
+      out.scalar_type() == torch::kBFloat16) {
+    launch_spmm<unsigned short, unsigned short>(
         (unsigned short*)out.data_ptr(), (const unsigned short*)x.data_ptr(),
         rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
         num_rows, D, accumulate, (size_t)x.numel(), stream);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    launch_spmm<unsigned short, float>(
+        out.data_ptr<float>(), (const unsigned short*)x.data_ptr(),
+        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
+        num_rows, D, accumulate, (size_t)x.numel(), stream);
   } else if (x.scalar_type() == torch::kFloat32) {
-    launch_spmm<float>(out.data_ptr<float>(), x.data_ptr<float>(),
+    launch_spmm<float, float>(out.data_ptr<float>(), x.data_ptr<float>(),
                        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd,
                        ds, ro, num_rows, D, accumulate, (size_t)x.numel(),
                        stream);

@@ -50,10 +50,23 @@ class _SpMM(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, rowptr, colidx, t_rowptr, t_colidx, num_rows, num_ext,
-                deg_dst, deg_src, row_order, t_row_order):
+                deg_dst, deg_src, row_order, t_row_order,
+                fwd_strips=None, bwd_strips=None):
         ctx.save_for_backward(t_rowptr, t_colidx, deg_dst, deg_src, t_row_order)
         ctx.num_ext = num_ext
-        if _hip(x):
+        ctx.bwd_strips = bwd_strips
+        if _hip(x) and fwd_strips is not None and deg_src is None:
+            # source-strip-blocked schedule: each pass gathers from an
+            # L2-resident source window; partials accumulate in fp32
+            # (one rounding total — same numerics as single-pass)
+            out32 = torch.empty(num_rows, x.shape[1], dtype=torch.float32,
+                                device=x.device)
+            for i, (srp, sci) in enumerate(fwd_strips):
+                _C.spmm(out32, x, srp, sci, None, None, None, i > 0)
+            if deg_dst is not None:
+                out32 = out32 * deg_dst.unsqueeze(1)
+            out = out32.to(x.dtype)
+        elif _hip(x):
             out = torch.empty(num_rows, x.shape[1], dtype=x.dtype, device=x.device)
             _C.spmm(out, x, rowptr, colidx, deg_dst, deg_src, row_order)
         else:
@@ -77,6 +90,12 @@ class _SpMM(torch.autograd.Function):
                 tmp = torch.empty_like(dy)
                 _C.rowscale(tmp, dy, deg_dst)
                 dy = tmp
+            if ctx.bwd_strips is not None and deg_src is None:
+                dx32 = torch.empty(ctx.num_ext, dy.shape[1],
+                                   dtype=torch.float32, device=dy.device)
+                for i, (srp, sci) in enumerate(ctx.bwd_strips):
+                    _C.spmm(dx32, dy, srp, sci, None, None, None, i > 0)
+                return (dx32.to(dy.dtype),) + (None,) * 12
             dx = torch.empty(ctx.num_ext, dy.shape[1], dtype=dy.dtype, device=dy.device)
             _C.spmm(dx, dy, t_rowptr, t_colidx, deg_src, None, t_row_order)
         else:
@@ -86,7 +105,7 @@ class _SpMM(torch.autograd.Function):
             dx = ref.spmm(yin, t_rowptr, t_colidx, ctx.num_ext)
             if deg_src is not None:
                 dx = dx * deg_src.unsqueeze(1).to(dx.dtype)
-        return (dx,) + (None,) * 10
+        return (dx,) + (None,) * 12
 
 
 class _SpMMEdge(torch.autograd.Function):
@@ -289,6 +308,7 @@ def scatter_gather(x, shard, normalize: bool = False, dst_scale=None,
         x, shard.rowptr, shard.colidx, shard.t_rowptr, shard.t_colidx,
         shard.n_local, shard.n_ext, deg_dst, deg_src,
         shard.row_order, shard.t_row_order,
+        shard.fwd_strips, shard.bwd_strips,
     )
 
 

@@ -98,13 +98,21 @@ class GraphShard:
     ag_self_row_order: Optional[torch.Tensor] = None  # degree-desc, int32
     ag_rem_row_order: Optional[torch.Tensor] = None
     ag_t_blk_order: Optional[torch.Tensor] = None  # [ws*mr] block-local
+    # --- source-strip-blocked SpMM plans (L2-resident gather windows;
+    # build_strip_plan; large local CSRs only) ------------------------
+    fwd_strips: Optional[list] = None  # [(rowptr, colidx), ...]
+    bwd_strips: Optional[list] = None  # same, for the transpose CSR
 
     def to(self, device) -> "GraphShard":
         d = {}
         for k, v in self.__dict__.items():
             if k.startswith("_"):  # lazy caches don't survive the move
                 continue
-            d[k] = v.to(device) if isinstance(v, torch.Tensor) else v
+            if isinstance(v, torch.Tensor):
+                v = v.to(device)
+            elif isinstance(v, list) and v and isinstance(v[0], tuple):
+                v = [tuple(t.to(device) for t in tup) for tup in v]
+            d[k] = v
         return GraphShard(**d)
 
     @property
@@ -150,6 +158,45 @@ class GraphShard:
             p = torch.argsort(self.colidx.long(), stable=True)
             object.__setattr__(self, "_t_eperm", p)
         return p
+
+
+def build_strip_plan(rowptr: torch.Tensor, colidx: torch.Tensor,
+                     num_cols: int, width: int):
+    """Split a CSR into ceil(num_cols/width) source-strip CSRs: strip s
+    holds each row's edges whose source id lies in [s*width,(s+1)*width).
+    Columns are sorted within rows, so every strip segment is a
+    contiguous row slice — one composite-key searchsorted per boundary.
+
+    Measured (profiles/r21): at width 16384 (≈8 MB of D=256 bf16 rows,
+    two XCD L2s) the multi-pass schedule runs the Reddit D=256 SpMM
+    1.39x faster than one pass — sources stay L2-resident per strip.
+    Partials accumulate in an fp32 buffer (mixed-output kernel), so
+    numerics are exactly one rounding, same as single-pass."""
+    rp = rowptr.numpy()
+    ci = colidx.numpy()
+    nrows = rp.shape[0] - 1
+    ci64 = ci.astype(np.int64)
+    row = np.repeat(np.arange(nrows, dtype=np.int64), np.diff(rp))
+    key = row * (num_cols + 1) + ci64
+    rows_q = np.arange(nrows, dtype=np.int64) * (num_cols + 1)
+    pos = [rp[:-1].astype(np.int64)]
+    for b in range(width, num_cols, width):
+        pos.append(np.searchsorted(key, rows_q + b))
+    pos.append(rp[1:].astype(np.int64))
+    strips = []
+    for s in range(len(pos) - 1):
+        seg_start, seg_end = pos[s], pos[s + 1]
+        cnt = seg_end - seg_start
+        srp = np.zeros(nrows + 1, dtype=np.int64)
+        np.cumsum(cnt, out=srp[1:])
+        if srp[-1]:
+            idx = np.repeat(seg_start - srp[:-1], cnt) + np.arange(
+                int(srp[-1]), dtype=np.int64)
+            sci = np.ascontiguousarray(ci[idx])
+        else:
+            sci = np.empty(0, dtype=ci.dtype)
+        strips.append((torch.from_numpy(srp), torch.from_numpy(sci)))
+    return strips
 
 
 def edge_tensor(shard: GraphShard, dim: Optional[int] = None,
@@ -485,6 +532,19 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
                                  shard_kw["halo_colidx"])
         shard_kw["t_loc_rowptr"], shard_kw["t_loc_colidx"] = t_loc
         shard_kw["t_halo_rowptr"], shard_kw["t_halo_colidx"] = t_halo
+
+    # source-strip-blocked plans for L2-resident gathers (measured
+    # 1.39x on the Reddit D=256 SpMM — profiles/r21). Only for CSRs big
+    # enough that the strip prep + extra output traffic pays.
+    strip_w = int(os.environ.get("ROC_SPMM_STRIP_WIDTH", "16384"))
+    strip_min = int(os.environ.get("ROC_SPMM_STRIP_MIN_EDGES",
+                                   str(24_000_000)))
+    if (strip_w > 0 and colidx_t.numel() >= strip_min
+            and n_ext > 2 * strip_w):
+        shard_kw["fwd_strips"] = build_strip_plan(rowptr_t, colidx_t,
+                                                  n_ext, strip_w)
+        shard_kw["bwd_strips"] = build_strip_plan(t_rowptr, t_colidx,
+                                                  n_local, strip_w)
 
     return GraphShard(
         rank=rank, world_size=world_size, bounds=list(bounds),
