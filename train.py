@@ -58,6 +58,8 @@ def parse_args():
                     help="sum replicates the reference's unscaled gradient")
     ap.add_argument("--eval-every", type=int, default=5)
     ap.add_argument("--residual", action="store_true")
+    ap.add_argument("--heads", type=int, default=4,
+                    help="GAT attention heads (concat; last layer 1)")
     ap.add_argument("--checkpoint", default=None)
     ap.add_argument("--checkpoint-every", type=int, default=0)
     ap.add_argument("--resume", default=None)
@@ -196,6 +198,8 @@ def main():
         shard = build_shard(g, rank, world, bounds,
                             use_comm=(world > 1 and dist.is_initialized()))
     mkw = {"residual": args.residual} if args.model == "gcn" else {}
+    if args.model == "gat":
+        mkw = {"heads": args.heads}
     model = build_model(args.model, dims, dropout=args.dropout,
                         seed=args.seed, **mkw)
     if args.recompute:
