@@ -539,8 +539,13 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
     strip_w = int(os.environ.get("ROC_SPMM_STRIP_WIDTH", "16384"))
     strip_min = int(os.environ.get("ROC_SPMM_STRIP_MIN_EDGES",
                                    str(24_000_000)))
+    max_k = int(os.environ.get("ROC_SPMM_STRIP_MAX_K", "32"))
+    # K passes cost K output re-reads: profitable only while K stays
+    # small AND each strip is L2-scale — huge graphs (papers100M:
+    # K would be ~6800) skip strips entirely
+    n_strips = (n_ext + strip_w - 1) // max(strip_w, 1)
     if (strip_w > 0 and colidx_t.numel() >= strip_min
-            and n_ext > 2 * strip_w):
+            and 2 <= n_strips <= max_k):
         shard_kw["fwd_strips"] = build_strip_plan(rowptr_t, colidx_t,
                                                   n_ext, strip_w)
         shard_kw["bwd_strips"] = build_strip_plan(t_rowptr, t_colidx,
