@@ -6,6 +6,8 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
           c10::optional<torch::Tensor> deg_src,
           c10::optional<torch::Tensor> row_order, bool accumulate);
 void rowscale(torch::Tensor out, torch::Tensor x, torch::Tensor scale);
+void cast_rowscale(torch::Tensor out, torch::Tensor x,
+                   c10::optional<torch::Tensor> scale);
 void relu_fwd(torch::Tensor out, torch::Tensor x);
 void sigmoid_fwd(torch::Tensor out, torch::Tensor x);
 void relu_bwd(torch::Tensor dx, torch::Tensor dy, torch::Tensor y);
@@ -54,6 +56,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("row_order") = pybind11::none(),
         pybind11::arg("accumulate") = false);
   m.def("rowscale", &rowscale);
+  m.def("cast_rowscale", &cast_rowscale,
+        "fp32 -> bf16 cast with optional row scale (strip epilogue)",
+        pybind11::arg("out"), pybind11::arg("x"),
+        pybind11::arg("scale") = pybind11::none());
   m.def("relu_fwd", &relu_fwd);
   m.def("sigmoid_fwd", &sigmoid_fwd);
   m.def("relu_bwd", &relu_bwd);

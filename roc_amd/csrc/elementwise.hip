@@ -154,3 +154,52 @@ void ewise_add(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
 void ewise_mul(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
   dispatch_map2(out, a, b, OpMul{});
 }
+
+// fp32 -> bf16 cast with optional per-row scale, one pass (the strip-
+// blocked SpMM epilogue: fp32 strip accumulator -> scaled bf16 out;
+// replaces a torch mul + a torch cast = 3 passes over the data).
+// Requires D % 8 == 0 (engine pads feature dims to 8).
+namespace {
+__global__ __launch_bounds__(kBlock) void cast_rowscale_kernel(
+    unsigned short* __restrict__ out, const float* __restrict__ x,
+    const float* __restrict__ scale, int64_t D, int64_t units) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t u = blockIdx.x * blockDim.x + threadIdx.x; u < units;
+       u += stride) {
+    const int64_t base = u * 8;
+    float a[8];
+    load_f32x4(x + base, a);
+    load_f32x4(x + base + 4, a + 4);
+    if (scale) {
+      const float s = scale[base / D];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) a[j] *= s;
+    }
+    store_bf16x8(out + base, a);
+  }
+}
+}  // namespace
+
+void cast_rowscale(torch::Tensor out, torch::Tensor x,
+                   c10::optional<torch::Tensor> scale) {
+  ROC_CHECK_DEV_CONT(out);
+  ROC_CHECK_DEV_CONT(x);
+  TORCH_CHECK(out.scalar_type() == torch::kBFloat16, "out must be bf16");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "x must be fp32");
+  TORCH_CHECK(out.sizes() == x.sizes(), "shape mismatch");
+  TORCH_CHECK(x.dim() == 2 && x.size(1) % 8 == 0,
+              "cast_rowscale needs D %% 8 == 0");
+  const float* sp = nullptr;
+  if (scale.has_value()) {
+    ROC_CHECK_DEV_CONT(*scale);
+    TORCH_CHECK(scale->scalar_type() == torch::kFloat32 &&
+                scale->numel() == x.size(0), "bad scale");
+    sp = scale->data_ptr<float>();
+  }
+  const int64_t units = x.numel() / 8;
+  const int grid = roc_grid_1d(units, kBlock, 2048);
+  hipLaunchKernelGGL(cast_rowscale_kernel, dim3(grid), dim3(kBlock), 0,
+                     roc_stream(), (unsigned short*)out.data_ptr(),
+                     x.data_ptr<float>(), sp, x.size(1), units);
+  ROC_HIP_CHECK(hipGetLastError());
+}

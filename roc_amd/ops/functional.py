@@ -63,9 +63,14 @@ class _SpMM(torch.autograd.Function):
                                 device=x.device)
             for i, (srp, sci) in enumerate(fwd_strips):
                 _C.spmm(out32, x, srp, sci, None, None, None, i > 0)
-            if deg_dst is not None:
-                out32 = out32 * deg_dst.unsqueeze(1)
-            out = out32.to(x.dtype)
+            if x.dtype == torch.bfloat16 and x.shape[1] % 8 == 0:
+                out = torch.empty(num_rows, x.shape[1],
+                                  dtype=torch.bfloat16, device=x.device)
+                _C.cast_rowscale(out, out32, deg_dst)  # fused epilogue
+            else:
+                if deg_dst is not None:
+                    out32 = out32 * deg_dst.unsqueeze(1)
+                out = out32.to(x.dtype)
         elif _hip(x):
             out = torch.empty(num_rows, x.shape[1], dtype=x.dtype, device=x.device)
             _C.spmm(out, x, rowptr, colidx, deg_dst, deg_src, row_order)
@@ -95,6 +100,12 @@ class _SpMM(torch.autograd.Function):
                                    dtype=torch.float32, device=dy.device)
                 for i, (srp, sci) in enumerate(ctx.bwd_strips):
                     _C.spmm(dx32, dy, srp, sci, None, None, None, i > 0)
+                if dy.dtype == torch.bfloat16 and dy.shape[1] % 8 == 0:
+                    dx = torch.empty(ctx.num_ext, dy.shape[1],
+                                     dtype=torch.bfloat16,
+                                     device=dy.device)
+                    _C.cast_rowscale(dx, dx32, None)
+                    return (dx,) + (None,) * 12
                 return (dx32.to(dy.dtype),) + (None,) * 12
             dx = torch.empty(ctx.num_ext, dy.shape[1], dtype=dy.dtype, device=dy.device)
             _C.spmm(dx, dy, t_rowptr, t_colidx, deg_src, None, t_row_order)
