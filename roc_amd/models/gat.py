@@ -71,16 +71,23 @@ class GAT(torch.nn.Module):
         h = F.dropout(x, self.p, self.training, call_id=i)
         h = F.linear(h, self.weights[i])            # [n_local, nh*dh]
         h_ext = halo_exchange(h, shard, group)      # [n_ext, nh*dh]
+        # ALL heads' score halves in one skinny GEMM each (a per-head
+        # fp32 gemv hit rocBLAS's slow gemvn path — 2.2 ms/call, 12% of
+        # the epoch, r2c16): block-diagonal [nh*dh, nh] projection,
+        # then transpose so each head's column is a contiguous row.
+        Wsrc = torch.block_diag(*(self.a_src[i][k].unsqueeze(1)
+                                  for k in range(nh))).to(h_ext.dtype)
+        Wdst = torch.block_diag(*(self.a_dst[i][k].unsqueeze(1)
+                                  for k in range(nh))).to(h_ext.dtype)
+        s_src_all = (h_ext @ Wsrc).float().t().contiguous()   # [nh, n_ext]
+        s_dst_all = (h_ext[:shard.n_local] @ Wdst).float().t().contiguous()
         outs = []
         for k in range(nh):
             hk = h_ext[:, k * dh:(k + 1) * dh]
-            # per-node halves of the additive score (fp32 for softmax)
-            s_src = (hk.float() @ self.a_src[i][k].float())   # [n_ext]
-            s_dst = (hk[:shard.n_local].float()
-                     @ self.a_dst[i][k].float())              # [n_local]
             # fused: lrelu(s_src[col]+s_dst[row]) -> segment softmax,
             # one kernel each direction (no E-length intermediates)
-            alpha = F.attention_softmax(s_src, s_dst, shard, self.slope)
+            alpha = F.attention_softmax(s_src_all[k], s_dst_all[k],
+                                        shard, self.slope)
             outs.append(F.scatter_gather_weighted(hk.contiguous(), alpha,
                                                   shard))
         out = outs[0] if nh == 1 else torch.cat(outs, dim=1)
