@@ -185,8 +185,35 @@ __global__ __launch_bounds__(kBlock) void att_softmax_bwd_kernel(
 #pragma unroll
     for (int off = TEAM / 2; off > 0; off >>= 1)
       dot += __shfl_xor(dot, off, 64);
+    // 4-deep unrolled so the random s_src gathers queue (the rolled
+    // version exposed one 4-B gather latency per edge — r2c16)
     float dd = 0.f;
-    for (int64_t e = e0 + lane; e < e1; e += TEAM) {
+    const int64_t span = e1 - e0;
+    const int64_t nfull = span / ((int64_t)TEAM * 4) * ((int64_t)TEAM * 4);
+    int64_t e = e0 + lane;
+    const int64_t estop = e0 + nfull;
+    for (; e < estop; e += (int64_t)TEAM * 4) {
+      const int64_t f0 = e, f1 = e + TEAM, f2 = e + 2 * TEAM,
+                    f3 = e + 3 * TEAM;
+      const int u0 = colidx[f0], u1 = colidx[f1];
+      const int u2 = colidx[f2], u3 = colidx[f3];
+      const float ss0 = s_src[u0], ss1 = s_src[u1];
+      const float ss2 = s_src[u2], ss3 = s_src[u3];
+      const float d0 = alpha[f0] * (dalpha[f0] - dot);
+      const float d1 = alpha[f1] * (dalpha[f1] - dot);
+      const float d2 = alpha[f2] * (dalpha[f2] - dot);
+      const float d3 = alpha[f3] * (dalpha[f3] - dot);
+      const float g0 = (ss0 + sd) > 0.f ? d0 : slope * d0;
+      const float g1 = (ss1 + sd) > 0.f ? d1 : slope * d1;
+      const float g2 = (ss2 + sd) > 0.f ? d2 : slope * d2;
+      const float g3 = (ss3 + sd) > 0.f ? d3 : slope * d3;
+      dd += g0 + g1 + g2 + g3;
+      atomicAdd(dsrc + u0, g0);
+      atomicAdd(dsrc + u1, g1);
+      atomicAdd(dsrc + u2, g2);
+      atomicAdd(dsrc + u3, g3);
+    }
+    for (; e < e1; e += TEAM) {
       const int u = colidx[e];
       const float ds = alpha[e] * (dalpha[e] - dot);
       const float raw = s_src[u] + sd;
