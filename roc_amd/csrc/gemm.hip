@@ -292,22 +292,98 @@ __global__ __launch_bounds__(kBlock) void gemm_atb_kernel(
   }
 }
 
+// 128x128 output-tile variant: 4x fewer cross-tile re-reads of A and B
+// (the 64x64 kernel re-read A per N-tile and B per Ka-tile — at
+// [608,256] that is 2.3 GB/call vs 1.16 GB here; measured r2c20).
+// Each wave owns 2 Ka row-groups (mi) x 8 N fragments (ni).
+template <typename ET, bool ALIGNED_A, bool ALIGNED_B>
+__global__ __launch_bounds__(kBlock) void gemm_atb_wide_kernel(
+    float* __restrict__ C, const ET* __restrict__ A, const ET* __restrict__ B,
+    int R, int Ka, int N, int rows_per_split) {
+  __shared__ ET at_lds[2 * BKA * PADK];
+  __shared__ ET bt_lds[2 * BNW * PADK];
+
+  const int i_blk = blockIdx.x * (2 * BKA);
+  const int n_blk = blockIdx.y * (2 * BNW);
+  const int r_begin = blockIdx.z * rows_per_split;
+  const int r_end = min(R, r_begin + rows_per_split);
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const int l15 = lane & 15;
+  const int khalf = lane >> 4;
+  const int i_wave = wave * 16;
+
+  f32x4 acc[2][8];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 8; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int r0 = r_begin; r0 < r_end; r0 += RB) {
+    stage_tile_T<ET, ALIGNED_A>(at_lds, A, r0, r_end, Ka, i_blk, Ka);
+    stage_tile_T<ET, ALIGNED_A>(at_lds + BKA * PADK, A, r0, r_end, Ka,
+                                i_blk + BKA, Ka);
+    stage_tile_T<ET, ALIGNED_B>(bt_lds, B, r0, r_end, N, n_blk, N);
+    stage_tile_T<ET, ALIGNED_B>(bt_lds + BNW * PADK, B, r0, r_end, N,
+                                n_blk + BNW, N);
+    __syncthreads();
+    const int irow = i_wave + l15;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < 8; ++ni) {
+        acc[mi][ni] = frag_mfma<ET>(at_lds + mi * BKA * PADK, bt_lds, irow,
+                                    ni * 16 + l15, khalf, acc[mi][ni]);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 8; ++ni) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int i = i_blk + mi * BKA + i_wave + khalf * 4 + j;
+        const int n = n_blk + ni * 16 + l15;
+        if (i < Ka && n < N && acc[mi][ni][j] != 0.f)
+          atomicAdd(&C[(int64_t)i * N + n], acc[mi][ni][j]);
+      }
+    }
+  }
+}
+
 template <typename ET>
 void launch_atb(float* c, const ET* a, const ET* b, int R, int Ka, int N,
                 hipStream_t s) {
-  const int tiles = ((Ka + BKA - 1) / BKA) * ((N + BNW - 1) / BNW);
+  const bool wide = Ka >= 2 * BKA && N >= 2 * BNW;
+  const int tw = wide ? 2 : 1;
+  const int tiles = ((Ka + tw * BKA - 1) / (tw * BKA)) *
+                    ((N + tw * BNW - 1) / (tw * BNW));
   int splitk = 2048 / max(tiles, 1);
   splitk = max(1, min(splitk, (R + RB - 1) / RB));
   const char* det = getenv("ROC_DETERMINISTIC");
   if (det && det[0] == '1') splitk = 1;  // bit-reproducible dW (slower)
   int rows_per_split = ((R + splitk - 1) / splitk + RB - 1) / RB * RB;
   splitk = (R + rows_per_split - 1) / rows_per_split;
-  dim3 grid((Ka + BKA - 1) / BKA, (N + BNW - 1) / BNW, splitk);
+  dim3 grid((Ka + tw * BKA - 1) / (tw * BKA),
+            (N + tw * BNW - 1) / (tw * BNW), splitk);
   const bool ala = (Ka % GemmTraits<ET>::kEPS) == 0;
   const bool alb = (N % GemmTraits<ET>::kEPS) == 0;
 #define ROC_ATB_CASE(ALA, ALB)                                                \
-  hipLaunchKernelGGL((gemm_atb_kernel<ET, ALA, ALB>), grid, dim3(kBlock), 0,  \
-                     s, c, a, b, R, Ka, N, rows_per_split)
+  do {                                                                        \
+    if (wide) {                                                               \
+      hipLaunchKernelGGL((gemm_atb_wide_kernel<ET, ALA, ALB>), grid,          \
+                         dim3(kBlock), 0, s, c, a, b, R, Ka, N,               \
+                         rows_per_split);                                     \
+    } else {                                                                  \
+      hipLaunchKernelGGL((gemm_atb_kernel<ET, ALA, ALB>), grid,               \
+                         dim3(kBlock), 0, s, c, a, b, R, Ka, N,               \
+                         rows_per_split);                                     \
+    }                                                                         \
+  } while (0)
   if (ala) { if (alb) ROC_ATB_CASE(true, true); else ROC_ATB_CASE(true, false); }
   else     { if (alb) ROC_ATB_CASE(false, true); else ROC_ATB_CASE(false, false); }
 #undef ROC_ATB_CASE

@@ -156,7 +156,9 @@ def test_gemm_rr_fused_relu():
 
 
 @pytest.mark.parametrize("R,Ka,N", [(5000, 608, 256), (3000, 256, 41),
-                                    (1000, 41, 64)])
+                                    (1000, 41, 64),
+                                    # wide 128x128-tile path, ragged edges
+                                    (2000, 200, 136), (1500, 136, 129)])
 def test_gemm_atb(R, Ka, N):
     torch.manual_seed(6)
     A = torch.randn(R, Ka).to(torch.bfloat16)
