@@ -229,3 +229,24 @@ def test_bench_reorder_script_cpu_dryrun():
         capture_output=True, text=True, timeout=300, cwd=repo)
     assert r.returncode == 0, r.stderr[-1000:]
     assert r.stdout.count("frac_in_64k_window") >= 3
+
+
+def test_predict_mode(tmp_path):
+    """--predict: checkpoint -> resume -> one infer forward -> .npy of
+    class ids (train.py inference mode; dropout=identity)."""
+    import numpy as np
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ck = str(tmp_path / "ck.pt")
+    out = str(tmp_path / "preds.npy")
+    base = [sys.executable, os.path.join(repo, "train.py"),
+            "--dataset", "cora-synthetic", "--scale", "0.2",
+            "--eval-every", "0"]
+    r = subprocess.run(base + ["--epochs", "3", "--checkpoint", ck],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    r = subprocess.run(base + ["--resume", ck, "--predict", out],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "[predict] wrote" in r.stdout
+    p = np.load(out)
+    assert p.dtype == np.int64 and p.min() >= 0 and p.max() < 7

@@ -65,6 +65,12 @@ def parse_args():
     ap.add_argument("--resume", default=None)
     ap.add_argument("--trace", default=None,
                     help="write a chrome trace JSON here at the end")
+    ap.add_argument("--predict", default=None,
+                    help="inference mode: skip training, run one forward "
+                         "in infer mode (dropout=identity) and write "
+                         "per-node predicted class ids (.npy, int64; "
+                         "multi-rank: one <path>.rankN.npy per rank with "
+                         "<path>.bounds.npy). Pair with --resume.")
     ap.add_argument("--offload", action="store_true",
                     help="host-DRAM activation offload (capacity tier)")
     ap.add_argument("--recompute", action="store_true",
@@ -264,6 +270,32 @@ def main():
 
             trainer.attach_windowed_dataset(lux, _loader,
                                             load_lux_meta(lux)[2])
+
+    if args.predict:
+        # inference mode: one forward with dropout=identity
+        # (reference infer_task semantics, `dropout_kernel.cu:159-180`)
+        import numpy as np
+        trainer.infer_mode()
+        with torch.no_grad():
+            logits = trainer.model(trainer.x, trainer.shard, trainer.group)
+            nc = num_classes or logits.shape[1]
+            pred = logits[:, :nc].float().argmax(dim=1).cpu().numpy()
+        if world > 1:
+            np.save(f"{args.predict}.rank{rank}.npy", pred)
+            if rank == 0:
+                np.save(f"{args.predict}.bounds.npy",
+                        np.asarray(shard.bounds))
+        else:
+            np.save(args.predict, pred)
+        md = trainer.evaluate()
+        if rank == 0:
+            print(f"[predict] wrote {args.predict} "
+                  f"({pred.shape[0]} local nodes); val_acc "
+                  f"{md['val_acc']:.4f} test_acc {md['test_acc']:.4f}",
+                  flush=True)
+        if world > 1:
+            dist.destroy_process_group()
+        return
 
     t_start = time.perf_counter()
     while trainer.epoch < args.epochs:
