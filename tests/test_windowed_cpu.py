@@ -114,3 +114,44 @@ def test_train_from_lux_two_ranks(tmp_path):
         cwd=repo, env=env, capture_output=True, text=True, timeout=420)
     assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1200:])
     assert "epoch     6" in r.stdout, r.stdout[-800:]
+
+
+def test_windowed_mean_loss_matches_single_rank(tmp_path):
+    """ADVICE-high regression: with windowed --file loading and the
+    default mean loss, every rank must use the GLOBAL train count for
+    the gradient scale — ws=2 training must produce the same weights
+    as the single-rank run on the same files (halo mode exercises the
+    gloo send-plan exchange too)."""
+    import subprocess
+    import sys
+    g = synthetic_graph(400, 6000, seed=9)
+    pref = str(tmp_path / "ds")
+    save_lux(pref + ".add_self_edge.lux", g)
+    rng = np.random.default_rng(3)
+    feats = rng.standard_normal((400, 8)).astype(np.float32)
+    feats.tofile(pref + ".feats.bin")
+    np.savetxt(pref + ".label", rng.integers(0, 3, 400), fmt="%d")
+    names = np.array(["Train", "Val", "Test", "None"])
+    with open(pref + ".mask", "w") as f:
+        f.write("\n".join(names[rng.integers(0, 4, 400)]) + "\n")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, ROC_COMM_MODE="halo")
+    base = [os.path.join(repo, "train.py"), "--file", pref, "--layers",
+            "8-16-3", "--epochs", "3", "--loss", "mean", "--dropout", "0",
+            "--eval-every", "0", "--seed", "1"]
+    r1 = subprocess.run([sys.executable] + base +
+                        ["--checkpoint", str(tmp_path / "ck1.pt")],
+                        capture_output=True, text=True, timeout=300, env=env)
+    assert r1.returncode == 0, r1.stderr[-800:]
+    r2 = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29741"] + base +
+        ["--checkpoint", str(tmp_path / "ck2.pt")],
+        capture_output=True, text=True, timeout=600, env=env, cwd=repo)
+    assert r2.returncode == 0, r2.stderr[-1500:]
+    s1 = torch.load(str(tmp_path / "ck1.pt"), weights_only=False)["model"]
+    s2 = torch.load(str(tmp_path / "ck2.pt"), weights_only=False)["model"]
+    for k in s1:
+        assert torch.allclose(s1[k], s2[k], atol=1e-5), \
+            (k, (s1[k] - s2[k]).abs().max())
