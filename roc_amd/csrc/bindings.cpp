@@ -4,7 +4,8 @@
 void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
           torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,
           c10::optional<torch::Tensor> deg_src,
-          c10::optional<torch::Tensor> row_order, bool accumulate);
+          c10::optional<torch::Tensor> row_order, bool accumulate,
+          int64_t col_base, int64_t ncols);
 void rowscale(torch::Tensor out, torch::Tensor x, torch::Tensor scale);
 void cast_rowscale(torch::Tensor out, torch::Tensor x,
                    c10::optional<torch::Tensor> scale);
@@ -54,7 +55,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("colidx"), pybind11::arg("deg_dst") = pybind11::none(),
         pybind11::arg("deg_src") = pybind11::none(),
         pybind11::arg("row_order") = pybind11::none(),
-        pybind11::arg("accumulate") = false);
+        pybind11::arg("accumulate") = false,
+        pybind11::arg("col_base") = 0, pybind11::arg("ncols") = 0);
   m.def("rowscale", &rowscale);
   m.def("cast_rowscale", &cast_rowscale,
         "fp32 -> bf16 cast with optional row scale (strip epilogue)",

@@ -181,16 +181,17 @@ __device__ __forceinline__ void row_accum_vec(
   }
 }
 
-// tail path: unit straddles D (only when D % EPU != 0; engine pads dims
-// to 8 so this is cold). Scalar loads, dynamic-bound loops.
+// tail path: unit straddles the column window (only when its width is
+// not a multiple of EPU; engine pads dims to 8 so this is cold).
+// Scalar loads, dynamic-bound loops.
 template <typename T, int EPU>
 __device__ void row_accum_tail(
-    float* __restrict__ acc, const T* __restrict__ x, int64_t D, int64_t col0,
+    float* __restrict__ acc, const T* __restrict__ x, int64_t ld, int64_t col0,
     int nvalid, const int* __restrict__ colidx,
     const float* __restrict__ deg_src, int64_t e0, int64_t e1) {
   for (int64_t e = e0; e < e1; ++e) {
     const int u0 = colidx[e];
-    const T* r = x + (int64_t)u0 * D + col0;
+    const T* r = x + (int64_t)u0 * ld + col0;
     const float w0 = deg_src ? deg_src[u0] : 1.f;
     for (int j = 0; j < nvalid; ++j) acc[j] += w0 * elt_to_f32(r[j]);
   }
@@ -202,22 +203,28 @@ __device__ void row_accum_tail(
 // OT (output type) may differ from T: OT=float with T=bf16 is the
 // strip-blocked accumulation path — fp32 partials in HBM, one rounding
 // at the final cast instead of one per strip pass.
+// ld is the row stride; [col_base, col_end) is the column window this
+// launch covers (ld == col_end, col_base == 0 for a whole-matrix pass;
+// a column-phase pass sweeps a 64-col window of a wider matrix so the
+// fp32 partial buffer is touched once per much-wider source strip).
 template <typename T, typename OT, int TEAM, int UN, bool BUF, bool SRC>
 __global__ __launch_bounds__(kBlock) void spmm_kernel(
     OT* __restrict__ out, const T* __restrict__ x,
     const int64_t* __restrict__ rowptr, const int* __restrict__ colidx,
     const float* __restrict__ deg_dst, const float* __restrict__ deg_src,
-    const int* __restrict__ row_order, int num_rows, int64_t D,
-    bool accumulate, unsigned x_bytes) {
+    const int* __restrict__ row_order, int num_rows, int64_t ld,
+    int64_t col_base, int64_t col_end, bool accumulate, unsigned x_bytes) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const int tpb = kBlock / TEAM;
   const int team = blockIdx.x * tpb + (int)threadIdx.x / TEAM;
   const int lane = (int)threadIdx.x % TEAM;
   const int nteams = gridDim.x * tpb;
 
-  const int64_t col0 = ((int64_t)blockIdx.y * TEAM + lane) * EPU;
-  const bool full = (col0 + EPU) <= D;
-  const int nvalid = full ? EPU : (col0 < D ? (int)(D - col0) : 0);
+  const int64_t col0 =
+      col_base + ((int64_t)blockIdx.y * TEAM + lane) * EPU;
+  const bool full = (col0 + EPU) <= col_end;
+  const int nvalid =
+      full ? EPU : (col0 < col_end ? (int)(col_end - col0) : 0);
 
   if (full) {
     for (int ri = team; ri < num_rows; ri += nteams) {
@@ -225,7 +232,7 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
       const int64_t e0 = rowptr[row];
       const int64_t e1 = rowptr[row + 1];
       float acc[EPU];
-      OT* o = out + (int64_t)row * D + col0;
+      OT* o = out + (int64_t)row * ld + col0;
       if (accumulate) {  // strip / halo-overlap pass 2: from partials
 #pragma unroll
         for (int j = 0; j < EPU; ++j) acc[j] = elt_to_f32(o[j]);
@@ -234,14 +241,14 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
         for (int j = 0; j < EPU; ++j) acc[j] = 0.f;
       }
       if constexpr (BUF) {
-        BufferGather<T> ld{
+        BufferGather<T> g{
             __builtin_amdgcn_make_buffer_rsrc((void*)x, (short)0, x_bytes,
                                               0x00020000),
-            (unsigned)(D * sizeof(T)), (unsigned)(col0 * sizeof(T))};
-        row_accum_vec<T, EPU, UN, SRC>(acc, ld, colidx, deg_src, e0, e1);
+            (unsigned)(ld * sizeof(T)), (unsigned)(col0 * sizeof(T))};
+        row_accum_vec<T, EPU, UN, SRC>(acc, g, colidx, deg_src, e0, e1);
       } else {
-        GlobalGather<T> ld{x, D, col0};
-        row_accum_vec<T, EPU, UN, SRC>(acc, ld, colidx, deg_src, e0, e1);
+        GlobalGather<T> g{x, ld, col0};
+        row_accum_vec<T, EPU, UN, SRC>(acc, g, colidx, deg_src, e0, e1);
       }
       if (deg_dst) {
         const float s = deg_dst[row];
@@ -260,10 +267,10 @@ __global__ __launch_bounds__(kBlock) void spmm_kernel(
     for (int ri = team; ri < num_rows; ri += nteams) {
       const int row = row_order ? row_order[ri] : ri;
       float acc[EPU];
-      OT* o = out + (int64_t)row * D + col0;
+      OT* o = out + (int64_t)row * ld + col0;
       for (int j = 0; j < nvalid; ++j)
         acc[j] = accumulate ? elt_to_f32(o[j]) : 0.f;
-      row_accum_tail<T, EPU>(acc, x, D, col0, nvalid, colidx, deg_src,
+      row_accum_tail<T, EPU>(acc, x, ld, col0, nvalid, colidx, deg_src,
                              rowptr[row], rowptr[row + 1]);
       const float s = deg_dst ? deg_dst[row] : 1.f;
       for (int j = 0; j < nvalid; ++j) f32_to_elt(acc[j] * s, o + j);
@@ -307,11 +314,12 @@ template <typename T, typename OT>
 void launch_spmm(OT* out, const T* x, const int64_t* rowptr,
                  const int* colidx, const float* deg_dst,
                  const float* deg_src, const int* row_order, int num_rows,
-                 int64_t D, bool accumulate, size_t x_elems,
-                 hipStream_t stream) {
+                 int64_t ld, int64_t col_base, int64_t col_end,
+                 bool accumulate, size_t x_elems, hipStream_t stream) {
   constexpr int EPU = EltTraits<T>::kPerVec;
   const SpmmKnobs& kn = spmm_knobs();
-  const int64_t units = (D + EPU - 1) / EPU;
+  const int64_t ncols = col_end - col_base;
+  const int64_t units = (ncols + EPU - 1) / EPU;
   int team = 8;
   while (team < units && team < 64) team *= 2;
   if (kn.team_override) team = kn.team_override;
@@ -332,14 +340,14 @@ void launch_spmm(OT* out, const T* x, const int64_t* rowptr,
       hipLaunchKernelGGL((spmm_kernel<T, OT, TEAM_, UN_, BUF_, true>),     \
                          grid,                                              \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
-                         deg_dst, deg_src, row_order, num_rows, D,          \
-                         accumulate, x_bytes);                              \
+                         deg_dst, deg_src, row_order, num_rows, ld,         \
+                         col_base, col_end, accumulate, x_bytes);           \
     } else {                                                                \
       hipLaunchKernelGGL((spmm_kernel<T, OT, TEAM_, UN_, BUF_, false>),    \
                          grid,                                              \
                          dim3(kBlock), 0, stream, out, x, rowptr, colidx,   \
-                         deg_dst, deg_src, row_order, num_rows, D,          \
-                         accumulate, x_bytes);                              \
+                         deg_dst, deg_src, row_order, num_rows, ld,         \
+                         col_base, col_end, accumulate, x_bytes);           \
     }                                                                       \
   } while (0)
 #define ROC_SPMM_L2(TEAM_)                                                  \
@@ -650,7 +658,8 @@ void spmm_refresh_knobs() { g_spmm_knobs_read = false; }
 void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
           torch::Tensor colidx, c10::optional<torch::Tensor> deg_dst,
           c10::optional<torch::Tensor> deg_src,
-          c10::optional<torch::Tensor> row_order, bool accumulate) {
+          c10::optional<torch::Tensor> row_order, bool accumulate,
+          int64_t col_base, int64_t ncols) {
   ROC_CHECK_DEV_CONT(out);
   ROC_CHECK_DEV_CONT(x);
   ROC_CHECK_DEV_CONT(rowptr);
@@ -665,6 +674,11 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
   const int64_t D = out.size(1);
   TORCH_CHECK(x.size(1) == D, "feature dim mismatch");
   TORCH_CHECK(rowptr.size(0) == num_rows + 1, "rowptr size mismatch");
+  // column-phase pass: touch only [col_base, col_base+ncols) of every
+  // row (ncols == 0 -> the whole row). The stride stays D.
+  const int64_t col_end = ncols > 0 ? col_base + ncols : D;
+  TORCH_CHECK(col_base >= 0 && col_base < col_end && col_end <= D,
+              "bad column window");
   const float* dd =
       deg_dst.has_value() ? deg_dst->data_ptr<float>() : nullptr;
   const float* ds =
@@ -677,17 +691,19 @@ void spmm(torch::Tensor out, torch::Tensor x, torch::Tensor rowptr,
     launch_spmm<unsigned short, unsigned short>(
         (unsigned short*)out.data_ptr(), (const unsigned short*)x.data_ptr(),
         rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
-        num_rows, D, accumulate, (size_t)x.numel(), stream);
+        num_rows, D, col_base, col_end, accumulate, (size_t)x.numel(),
+        stream);
   } else if (x.scalar_type() == torch::kBFloat16) {
     launch_spmm<unsigned short, float>(
         out.data_ptr<float>(), (const unsigned short*)x.data_ptr(),
         rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd, ds, ro,
-        num_rows, D, accumulate, (size_t)x.numel(), stream);
+        num_rows, D, col_base, col_end, accumulate, (size_t)x.numel(),
+        stream);
   } else if (x.scalar_type() == torch::kFloat32) {
     launch_spmm<float, float>(out.data_ptr<float>(), x.data_ptr<float>(),
                        rowptr.data_ptr<int64_t>(), colidx.data_ptr<int>(), dd,
-                       ds, ro, num_rows, D, accumulate, (size_t)x.numel(),
-                       stream);
+                       ds, ro, num_rows, D, col_base, col_end, accumulate,
+                       (size_t)x.numel(), stream);
   } else {
     TORCH_CHECK(false, "spmm: unsupported dtype (bf16/f32 only)");
   }

@@ -27,6 +27,23 @@ def has_ext() -> bool:
     return _C is not None
 
 
+_PHASE_COLS: Optional[int] = None
+
+
+def _phase_cols() -> int:
+    """Column-phase width for strip-blocked SpMM passes (0 = off).
+
+    Pinned at first use for the same reason the kernel geometry knobs
+    are: a per-launch getenv on the hot path silently desyncs from any
+    hipGraph capture. 64 cols x 8 lanes covers one TEAM=8 unit exactly.
+    """
+    global _PHASE_COLS
+    if _PHASE_COLS is None:
+        import os
+        _PHASE_COLS = int(os.environ.get("ROC_SPMM_PHASE_COLS", "64"))
+    return _PHASE_COLS
+
+
 def _hip(x: torch.Tensor) -> bool:
     if x.is_cuda:
         if _C is None:
@@ -58,11 +75,24 @@ class _SpMM(torch.autograd.Function):
         if _hip(x) and fwd_strips is not None and deg_src is None:
             # source-strip-blocked schedule: each pass gathers from an
             # L2-resident source window; partials accumulate in fp32
-            # (one rounding total — same numerics as single-pass)
+            # (one rounding total — same numerics as single-pass).
+            # Wide feature dims additionally sweep column PHASES: a
+            # 64-col phase over 4x-wider strips keeps the same gather
+            # working set (rows x cols bytes) while the fp32 partial
+            # buffer is re-read/re-written 4x less often — partial
+            # traffic, not gathers, was the next bound (profiles/r23).
             out32 = torch.empty(num_rows, x.shape[1], dtype=torch.float32,
                                 device=x.device)
-            for i, (srp, sci) in enumerate(fwd_strips):
-                _C.spmm(out32, x, srp, sci, None, None, None, i > 0)
+            D = x.shape[1]
+            ph = _phase_cols()
+            if ph and D % ph == 0 and D // ph >= 2:
+                for p in range(0, D, ph):
+                    for i, (srp, sci) in enumerate(fwd_strips):
+                        _C.spmm(out32, x, srp, sci, None, None, None,
+                                i > 0, p, ph)
+            else:
+                for i, (srp, sci) in enumerate(fwd_strips):
+                    _C.spmm(out32, x, srp, sci, None, None, None, i > 0)
             if x.dtype == torch.bfloat16 and x.shape[1] % 8 == 0:
                 out = torch.empty(num_rows, x.shape[1],
                                   dtype=torch.bfloat16, device=x.device)
@@ -98,8 +128,16 @@ class _SpMM(torch.autograd.Function):
             if ctx.bwd_strips is not None and deg_src is None:
                 dx32 = torch.empty(ctx.num_ext, dy.shape[1],
                                    dtype=torch.float32, device=dy.device)
-                for i, (srp, sci) in enumerate(ctx.bwd_strips):
-                    _C.spmm(dx32, dy, srp, sci, None, None, None, i > 0)
+                D = dy.shape[1]
+                ph = _phase_cols()
+                if ph and D % ph == 0 and D // ph >= 2:
+                    for p in range(0, D, ph):
+                        for i, (srp, sci) in enumerate(ctx.bwd_strips):
+                            _C.spmm(dx32, dy, srp, sci, None, None, None,
+                                    i > 0, p, ph)
+                else:
+                    for i, (srp, sci) in enumerate(ctx.bwd_strips):
+                        _C.spmm(dx32, dy, srp, sci, None, None, None, i > 0)
                 if dy.dtype == torch.bfloat16 and dy.shape[1] % 8 == 0:
                     dx = torch.empty(ctx.num_ext, dy.shape[1],
                                      dtype=torch.bfloat16,

@@ -338,6 +338,48 @@ def test_spmm_strips_gpu(monkeypatch):
                               rtol=0.02), dtype
 
 
+def test_spmm_col_phases_gpu(monkeypatch):
+    """Column-phase strip passes (64-col windows over wide strips) must
+    match whole-row strip passes AND the single-pass kernel bit-for-bit
+    in fp32 (all three accumulate in source-ascending order)."""
+    import roc_amd.ops.functional as Fn
+    from roc_amd import build_shard
+    from roc_amd.graph import synthetic_graph
+    torch.manual_seed(17)
+    g = synthetic_graph(1500, 60000, seed=11)
+    sh_plain = build_shard(g, 0, 1).to(DEV)
+    monkeypatch.setenv("ROC_SPMM_STRIP_MIN_EDGES", "0")
+    monkeypatch.setenv("ROC_SPMM_STRIP_WIDTH", "256")
+    sh_strip = build_shard(g, 0, 1).to(DEV)
+    assert sh_strip.fwd_strips is not None
+    for dtype in (torch.float32, torch.bfloat16):
+        for D in (128, 192):  # phase applies: D % 64 == 0, >= 2 phases
+            x = torch.randn(g.num_nodes, D, device=DEV).to(dtype)
+            gy = torch.randn(g.num_nodes, D, device=DEV).to(dtype)
+            outs, grads = [], []
+            for ph in (0, 64):
+                monkeypatch.setattr(Fn, "_PHASE_COLS", ph)
+                xg = x.clone().requires_grad_(True)
+                y = F.scatter_gather(xg, sh_strip,
+                                     dst_scale=sh_strip.rsqrt_deg_local)
+                y.backward(gy)
+                outs.append(y.float())
+                grads.append(xg.grad.float())
+            xg = x.clone().requires_grad_(True)
+            y = F.scatter_gather(xg, sh_plain,
+                                 dst_scale=sh_plain.rsqrt_deg_local)
+            y.backward(gy)
+            outs.append(y.float())
+            grads.append(xg.grad.float())
+            monkeypatch.setattr(Fn, "_PHASE_COLS", None)
+            for o in outs[1:]:
+                assert torch.equal(outs[0], o) if dtype == torch.float32 \
+                    else torch.allclose(outs[0], o, atol=1e-2, rtol=0.02)
+            for gr in grads[1:]:
+                assert torch.equal(grads[0], gr) if dtype == torch.float32 \
+                    else torch.allclose(grads[0], gr, atol=1e-2, rtol=0.02)
+
+
 def test_edge_softmax_gpu():
     """Fused segment-softmax kernel fwd+bwd vs the CPU reference."""
     from roc_amd import build_shard
