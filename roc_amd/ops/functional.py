@@ -33,14 +33,19 @@ _PHASE_COLS: Optional[int] = None
 def _phase_cols() -> int:
     """Column-phase width for strip-blocked SpMM passes (0 = off).
 
+    Default OFF by measurement (profiles/r23): at equal gather
+    working-set bytes, 64-col phases run the headline epoch 19.4 ms vs
+    16.0 ms single-phase — narrow 128-B row gathers pay a ~20% service
+    penalty vs whole 512-B rows, and the partial-traffic saving never
+    lands because phases multiply the pass count right back.
     Pinned at first use for the same reason the kernel geometry knobs
     are: a per-launch getenv on the hot path silently desyncs from any
-    hipGraph capture. 64 cols x 8 lanes covers one TEAM=8 unit exactly.
+    hipGraph capture.
     """
     global _PHASE_COLS
     if _PHASE_COLS is None:
         import os
-        _PHASE_COLS = int(os.environ.get("ROC_SPMM_PHASE_COLS", "64"))
+        _PHASE_COLS = int(os.environ.get("ROC_SPMM_PHASE_COLS", "0"))
     return _PHASE_COLS
 
 

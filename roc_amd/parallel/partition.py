@@ -536,10 +536,10 @@ def build_shard_from_window(rowptr_full: torch.Tensor,
     # source-strip-blocked plans for L2-resident gathers (measured
     # 1.39x on the Reddit D=256 SpMM — profiles/r21). Only for CSRs big
     # enough that the strip prep + extra output traffic pays.
-    # With column phases on (the default), passes touch 64-col windows,
-    # so strips can be 4x wider at the same gather working set — fewer
-    # passes over the fp32 partial buffer (see _SpMM.forward).
-    phase_on = int(os.environ.get("ROC_SPMM_PHASE_COLS", "64")) > 0
+    # With column phases on (off by default — see _phase_cols()),
+    # passes touch 64-col windows, so strips are 4x wider at the same
+    # gather working set.
+    phase_on = int(os.environ.get("ROC_SPMM_PHASE_COLS", "0")) > 0
     strip_w = int(os.environ.get("ROC_SPMM_STRIP_WIDTH",
                                  "65536" if phase_on else "16384"))
     strip_min = int(os.environ.get("ROC_SPMM_STRIP_MIN_EDGES",
