@@ -59,14 +59,23 @@ def build_strips(rowptr, colidx, n, width):
 
 
 def main():
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--dim", type=int, default=256)
+    ap.add_argument("--widths", default="65536,32768,16384,8192")
+    ap.add_argument("--fp32acc", action="store_true",
+                    help="production path: fp32 partial buffer + final "
+                         "cast_rowscale instead of bf16 in-place passes")
+    args = ap.parse_args()
     dev = "cuda:0"
     torch.manual_seed(0)
     g, feats, *_ = synthetic_dataset("reddit", seed=1)
     sh = build_shard(g, 0, 1)
     n = g.num_nodes
-    D = 256
+    D = args.dim
     x = torch.randn(n, D, device=dev).to(torch.bfloat16)
     out = torch.empty(n, D, device=dev, dtype=torch.bfloat16)
+    out32 = torch.empty(n, D, device=dev, dtype=torch.float32)
     rp_d = sh.rowptr.to(dev)
     ci_d = sh.colidx.to(dev)
 
@@ -84,16 +93,22 @@ def main():
     print(f"single-pass D={D}: {base:.2f} ms", flush=True)
     ref = out.float().cpu().clone()
 
-    for width in (65536, 32768, 16384, 8192):
+    for width in (int(w) for w in args.widths.split(",")):
         t0 = time.perf_counter()
         strips = [(srp.to(dev), sci.to(dev))
                   for srp, sci in build_strips(sh.rowptr, sh.colidx, n,
                                                width)]
         prep = time.perf_counter() - t0
 
-        def run():
-            for i, (srp, sci) in enumerate(strips):
-                _C.spmm(out, x, srp, sci, None, None, None, i > 0)
+        if args.fp32acc:
+            def run():
+                for i, (srp, sci) in enumerate(strips):
+                    _C.spmm(out32, x, srp, sci, None, None, None, i > 0)
+                _C.cast_rowscale(out, out32, None)
+        else:
+            def run():
+                for i, (srp, sci) in enumerate(strips):
+                    _C.spmm(out, x, srp, sci, None, None, None, i > 0)
 
         t = timeit(run)
         err = (out.float().cpu() - ref).abs().max().item()
