@@ -362,7 +362,14 @@ void launch_atb(float* c, const ET* a, const ET* b, int R, int Ka, int N,
   const int tw = wide ? 2 : 1;
   const int tiles = ((Ka + tw * BKA - 1) / (tw * BKA)) *
                     ((N + tw * BNW - 1) / (tw * BNW));
-  int splitk = 2048 / max(tiles, 1);
+  // split-K count trades fill (tiles*splitk workgroups) against the
+  // atomic RMW volume into C (Ka*N*splitk adds); pinned once at first
+  // launch (same hipGraph-capture rule as the SpMM knobs)
+  static const int splitk_env = [] {
+    const char* e = getenv("ROC_ATB_SPLITK");
+    return e ? atoi(e) : 0;
+  }();
+  int splitk = splitk_env > 0 ? splitk_env : 2048 / max(tiles, 1);
   splitk = max(1, min(splitk, (R + RB - 1) / RB));
   const char* det = getenv("ROC_DETERMINISTIC");
   if (det && det[0] == '1') splitk = 1;  // bit-reproducible dW (slower)
