@@ -46,14 +46,25 @@ class GCN(torch.nn.Module):
         w = self.weights[i]
         h = F.dropout(x, self.p, self.training, call_id=i)
         if self.fused_norm:
-            # source-side D^-1/2 rides the GEMM epilogue (owner rank
-            # scales its rows BEFORE the halo exchange); dst-side
-            # D^-1/2 rides the SpMM store. No per-edge degree gather.
-            h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
-            # strategy (halo a2a / overlap / all_gather) is the
-            # shard's choice — see parallel/aggregate.py
-            h = aggregate(h, shard, dst_scale=shard.rsqrt_deg_local,
-                          group=group)
+            if w.shape[0] < w.shape[1]:
+                # widening layer: aggregate FIRST, at the narrower
+                # input width (aggregation is linear: A(XW) == (AX)W),
+                # so the gather stream AND any halo exchange move
+                # in_dim-wide rows instead of out_dim-wide ones
+                h = F.degree_scale(h, shard.rsqrt_deg_local)
+                h = aggregate(h, shard, dst_scale=shard.rsqrt_deg_local,
+                              group=group)
+                h = F.linear(h, w)
+            else:
+                # source-side D^-1/2 rides the GEMM epilogue (owner
+                # rank scales its rows BEFORE the halo exchange);
+                # dst-side D^-1/2 rides the SpMM store. No per-edge
+                # degree gather.
+                h = F.linear(h, w, row_scale=shard.rsqrt_deg_local)
+                # strategy (halo a2a / overlap / all_gather) is the
+                # shard's choice — see parallel/aggregate.py
+                h = aggregate(h, shard, dst_scale=shard.rsqrt_deg_local,
+                              group=group)
         else:
             h = F.linear(h, w)
             h = halo_exchange(h, shard, group)

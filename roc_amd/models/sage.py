@@ -3,9 +3,11 @@
 h' = relu( W_self·h_v  +  W_neigh·mean_{u in N(v)} h_u )
 
 The concat-then-linear of the original formulation is algebraically folded
-into two GEMMs (no concat materialization), and W_neigh is applied BEFORE
-the halo exchange + aggregation so the exchanged rows are already at the
-(usually smaller) output width.
+into two GEMMs (no concat materialization), and the aggregation runs at
+the NARROWER of (input, output) width — W_neigh is applied before the
+exchange+aggregation on shrinking layers and after them on widening
+ones (mean_u(h W) == (mean_u h) W) — so the gather stream and the halo
+exchange always move the skinnier rows.
 """
 from __future__ import annotations
 
@@ -44,9 +46,18 @@ class GraphSAGE(torch.nn.Module):
     def _layer(self, i, x, shard, group):
         h = F.dropout(x, self.p, self.training, call_id=i)
         h_self = F.linear(h, self.w_self[i])
-        hn = F.linear(h, self.w_neigh[i])
-        hn = aggregate(hn, shard, dst_scale=shard.inv_deg_local,
-                       group=group)  # fused mean
+        wn = self.w_neigh[i]
+        if wn.shape[0] < wn.shape[1]:
+            # widening layer: mean-aggregate first at the narrower
+            # input width (mean_u(h W) == (mean_u h) W) — smaller
+            # gather stream and halo exchange
+            hn = aggregate(h, shard, dst_scale=shard.inv_deg_local,
+                           group=group)
+            hn = F.linear(hn, wn)
+        else:
+            hn = F.linear(h, wn)
+            hn = aggregate(hn, shard, dst_scale=shard.inv_deg_local,
+                           group=group)  # fused mean
         h = F.add(h_self, hn)
         if i < len(self.w_self) - 1:
             h = F.relu(h)
