@@ -9,12 +9,26 @@ kernel (`scatter_gather(..., normalize=True)`).
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from ..ops import functional as F
 from ..ops.reference import glorot_uniform
 from ..parallel.halo import halo_exchange
 from ..parallel.aggregate import aggregate
+
+
+def _adaptive_agg() -> bool:
+    """Aggregate-at-narrower-width on widening layers (ROC_ADAPTIVE_AGG,
+    default on; 0 disables for A/B)."""
+    global _ADAPTIVE
+    if _ADAPTIVE is None:
+        _ADAPTIVE = os.environ.get("ROC_ADAPTIVE_AGG", "1") != "0"
+    return _ADAPTIVE
+
+
+_ADAPTIVE = None
 
 
 class GCN(torch.nn.Module):
@@ -46,7 +60,7 @@ class GCN(torch.nn.Module):
         w = self.weights[i]
         h = F.dropout(x, self.p, self.training, call_id=i)
         if self.fused_norm:
-            if w.shape[0] < w.shape[1]:
+            if w.shape[0] < w.shape[1] and _adaptive_agg():
                 # widening layer: aggregate FIRST, at the narrower
                 # input width (aggregation is linear: A(XW) == (AX)W),
                 # so the gather stream AND any halo exchange move

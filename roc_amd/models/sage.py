@@ -47,7 +47,8 @@ class GraphSAGE(torch.nn.Module):
         h = F.dropout(x, self.p, self.training, call_id=i)
         h_self = F.linear(h, self.w_self[i])
         wn = self.w_neigh[i]
-        if wn.shape[0] < wn.shape[1]:
+        from .gcn import _adaptive_agg
+        if wn.shape[0] < wn.shape[1] and _adaptive_agg():
             # widening layer: mean-aggregate first at the narrower
             # input width (mean_u(h W) == (mean_u h) W) — smaller
             # gather stream and halo exchange
