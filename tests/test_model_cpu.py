@@ -145,3 +145,23 @@ def test_recompute_matches_standard_gradients():
         assert len(g0) == len(g1) and len(g0) > 0
         for a, b in zip(g0, g1):
             assert torch.allclose(a, b, rtol=1e-5, atol=1e-6), name
+
+
+@pytest.mark.parametrize("name", ["gcn", "sage"])
+def test_adaptive_agg_order_equal(name, monkeypatch):
+    """Aggregate-first (widening layers) must equal post-GEMM
+    aggregation: A(XW) == (AX)W up to fp rounding. Uses a widening
+    hidden layer (in < hidden) so the flip actually engages."""
+    import roc_amd.models.gcn as gcn_mod
+    g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.05)
+    shard = build_shard(g, 0, 1)
+    dims = [feats.shape[1], feats.shape[1] * 2, c]  # layer 1 widens
+    outs = []
+    for flag in (True, False):
+        monkeypatch.setattr(gcn_mod, "_ADAPTIVE", flag)
+        m = build_model(name, dims, dropout=0.0, seed=1)
+        m.eval()
+        outs.append(m(feats.float(), shard))
+    monkeypatch.setattr(gcn_mod, "_ADAPTIVE", None)
+    assert torch.allclose(outs[0], outs[1], atol=1e-4, rtol=1e-4), \
+        (outs[0] - outs[1]).abs().max()
