@@ -315,3 +315,64 @@ def test_elastic_resume_ws1_to_ws2(tmp_path):
         assert err is None, f"rank {rank}: {err}"
     assert np.allclose(res[0][1], res[1][1], atol=1e-6)
     assert np.allclose(res[0][1], w_ref, atol=2e-3)  # Adam noise tolerance
+
+
+def _widen_worker(rank, port, mode, model_name, q):
+    # widening hidden layer (in < hidden): the adaptive aggregation
+    # order flips to aggregate-FIRST, so the halo/allgather exchange
+    # moves the in-width rows — this worker covers that comm path
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["ROC_COMM_MODE"] = mode
+        dist.init_process_group("gloo", rank=rank, world_size=WS)
+        torch.manual_seed(0)
+        g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.05,
+                                                      seed=3)
+        bounds = edge_balanced_bounds(g.rowptr, WS)
+        sh = build_shard(g, rank, WS, bounds)
+        dims = [feats.shape[1], feats.shape[1] * 2, c]
+        model = build_model(model_name, dims, dropout=0.0, seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
+        tr = Trainer(model, sh, feats, labels, mask, opt)
+        for _ in range(3):
+            tr.train_epoch()
+        w = next(model.parameters()).detach().numpy().copy()
+        q.put((rank, w, None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, None, repr(e)))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("mode,model_name,port", [
+    ("halo", "gcn", 29531), ("allgather", "gcn", 29533),
+    ("halo", "sage", 29535),
+])
+def test_widening_layer_sharded_matches_single(mode, model_name, port):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_widen_worker,
+                         args=(r, port, mode, model_name, q))
+             for r in range(WS)]
+    for p in procs:
+        p.start()
+    res = sorted([q.get() for _ in range(WS)], key=lambda t: t[0])
+    for p in procs:
+        p.join(timeout=180)
+    for rank, w, err in res:
+        assert err is None, f"rank {rank}: {err}"
+    assert np.allclose(res[0][1], res[1][1], atol=1e-6)
+
+    torch.manual_seed(0)
+    g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.05, seed=3)
+    sh = build_shard(g, 0, 1)
+    dims = [feats.shape[1], feats.shape[1] * 2, c]
+    model = build_model(model_name, dims, dropout=0.0, seed=1)
+    opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
+    tr = Trainer(model, sh, feats, labels, mask, opt)
+    for _ in range(3):
+        tr.train_epoch()
+    w_single = next(model.parameters()).detach().numpy()
+    assert np.allclose(res[0][1], w_single, atol=2e-3)
