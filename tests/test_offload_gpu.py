@@ -96,3 +96,24 @@ def test_offload_declares_stream_edges():
         assert sc.pending() == []
     finally:
         sc.enable_stream_debug(False)
+
+
+def test_streamcheck_live_on_gpu_offload():
+    """Race-detection debug mode on REAL device streams: with stream
+    debug enabled, an offloaded training epoch must validate every
+    cross-stream edge (producer registered before each consumer, no
+    violation raised) and log the full offload edge chain."""
+    from roc_amd import streamcheck
+    streamcheck.enable_stream_debug(True)
+    try:
+        tr = _make(offload=True)
+        for _ in range(2):
+            tr.train_epoch()
+        torch.cuda.synchronize()
+        log = streamcheck.edge_log()
+        prods = {p for p, _ in log}
+        assert "offload-d2h" in prods, log[:8]
+        assert "offload-h2d" in prods, log[:8]
+        assert any(c == "backward-consume" for _, c in log)
+    finally:
+        streamcheck.enable_stream_debug(False)
