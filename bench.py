@@ -211,7 +211,8 @@ def main():
             "higher_is_better": False,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": args.dtype,
+            # actual compute dtype (CPU dry runs force fp32)
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
             "data": "synthetic",
             "config": {
                 "model": f"{args.model}-{args.layers}layer-" +
