@@ -217,10 +217,23 @@ constexpr int BNW = 64;  // N tile (output cols)
 constexpr int RB = 32;   // reduction rows per step
 
 // stage a [RB x 64] tile TRANSPOSED into LDS[64][PADK] (lds[c][r]).
+//
+// bf16 swizzle: the plain [c][r] image puts every lane of a half-wave
+// on the same bank pair (column stride 64*PADK bytes for a fixed r is
+// 0 mod the 32 write banks), measured at 22% of gemm_atb_wide's wave
+// cycles in SQ_LDS_BANK_CONFLICT (profiles/r29). Storing row
+// r ^ ((cseg&3)<<3) spreads the 8 column-segment lanes over 4 bank
+// groups while keeping each khalf-read's 8 elements contiguous; the
+// reader compensates with khalf ^ ((c>>3)&3) (frag_mfma_tswz).
+// LOAD half: globals -> regs (issue-only; the s_waitcnt lands in the
+// WRITE half, so callers can put a full MFMA phase between the two —
+// the T14 split; the fused version had zero latency hiding)
 template <typename ET, bool ALIGNED>
-__device__ __forceinline__ void stage_tile_T(
-    ET* __restrict__ lds, const ET* __restrict__ g, int r0, int nrows,
-    int64_t ld, int c0, int ncols) {
+__device__ __forceinline__ void stage_loadT(
+    typename GemmTraits<ET>::Seg (&regs)[RB * 64 / kBlock /
+                                         GemmTraits<ET>::kEPS],
+    const ET* __restrict__ g, int r0, int nrows, int64_t ld, int c0,
+    int ncols) {
   constexpr int EPS = GemmTraits<ET>::kEPS;
   constexpr int SPR = 64 / EPS;           // segments per row of 64 cols
   constexpr int RPP = kBlock / SPR;       // rows per pass
@@ -240,8 +253,57 @@ __device__ __forceinline__ void stage_tile_T(
         for (int j = 0; j < max(nv, 0); ++j) v[j] = p[j];
       }
     }
+    regs[pass] = v;
+  }
+}
+
+// WRITE half: regs -> transposed (bf16: bank-swizzled) LDS image
+template <typename ET>
+__device__ __forceinline__ void stage_writeT(
+    ET* __restrict__ lds,
+    const typename GemmTraits<ET>::Seg (&regs)[RB * 64 / kBlock /
+                                               GemmTraits<ET>::kEPS]) {
+  constexpr int EPS = GemmTraits<ET>::kEPS;
+  constexpr int SPR = 64 / EPS;
+  constexpr int RPP = kBlock / SPR;
+  const int cseg = threadIdx.x % SPR;
+  const int r_in = threadIdx.x / SPR;
 #pragma unroll
-    for (int j = 0; j < EPS; ++j) lds[(cseg * EPS + j) * PADK + r] = (ET)v[j];
+  for (int pass = 0; pass < RB / RPP; ++pass) {
+    const int r = pass * RPP + r_in;
+    const int rs = (sizeof(ET) == 2) ? (r ^ ((cseg & 3) << 3)) : r;
+#pragma unroll
+    for (int j = 0; j < EPS; ++j)
+      lds[(cseg * EPS + j) * PADK + rs] = (ET)regs[pass][j];
+  }
+}
+
+template <typename ET, bool ALIGNED>
+__device__ __forceinline__ void stage_tile_T(
+    ET* __restrict__ lds, const ET* __restrict__ g, int r0, int nrows,
+    int64_t ld, int c0, int ncols) {
+  typename GemmTraits<ET>::Seg regs[RB * 64 / kBlock /
+                                    GemmTraits<ET>::kEPS];
+  stage_loadT<ET, ALIGNED>(regs, g, r0, nrows, ld, c0, ncols);
+  stage_writeT<ET>(lds, regs);
+}
+
+// fragment reader for the swizzled transposed images (bf16; the fp32
+// image is stored unswizzled and read by the generic frag_mfma)
+template <typename ET>
+__device__ __forceinline__ f32x4 frag_mfma_tswz(
+    const ET* a_lds, const ET* b_lds, int mrow, int nrow, int khalf,
+    f32x4 acc) {
+  if constexpr (sizeof(ET) == 2) {
+    const int ka = (khalf ^ ((mrow >> 3) & 3)) * 8;
+    const int kb = (khalf ^ ((nrow >> 3) & 3)) * 8;
+    const short8 af =
+        *reinterpret_cast<const short8*>(&a_lds[mrow * PADK + ka]);
+    const short8 bf =
+        *reinterpret_cast<const short8*>(&b_lds[nrow * PADK + kb]);
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+  } else {
+    return frag_mfma<ET>(a_lds, b_lds, mrow, nrow, khalf, acc);
   }
 }
 
@@ -267,15 +329,26 @@ __global__ __launch_bounds__(kBlock) void gemm_atb_kernel(
 #pragma unroll
   for (int j = 0; j < 4; ++j) acc[j] = {0.f, 0.f, 0.f, 0.f};
 
+  // T14 split pipeline: chunk r0+RB's global loads are issued right
+  // after the barrier, a full MFMA phase before their s_waitcnt in
+  // stage_writeT (the fused stage had zero latency hiding)
+  typename GemmTraits<ET>::Seg ra[RB * 64 / kBlock / GemmTraits<ET>::kEPS];
+  typename GemmTraits<ET>::Seg rb[RB * 64 / kBlock / GemmTraits<ET>::kEPS];
+  stage_loadT<ET, ALIGNED_A>(ra, A, r_begin, r_end, Ka, i_blk, Ka);
+  stage_loadT<ET, ALIGNED_B>(rb, B, r_begin, r_end, N, n_blk, N);
   for (int r0 = r_begin; r0 < r_end; r0 += RB) {
-    stage_tile_T<ET, ALIGNED_A>(at_lds, A, r0, r_end, Ka, i_blk, Ka);
-    stage_tile_T<ET, ALIGNED_B>(bt_lds, B, r0, r_end, N, n_blk, N);
+    stage_writeT<ET>(at_lds, ra);
+    stage_writeT<ET>(bt_lds, rb);
     __syncthreads();
+    if (r0 + RB < r_end) {
+      stage_loadT<ET, ALIGNED_A>(ra, A, r0 + RB, r_end, Ka, i_blk, Ka);
+      stage_loadT<ET, ALIGNED_B>(rb, B, r0 + RB, r_end, N, n_blk, N);
+    }
     const int irow = i_wave + l15;
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
-      acc[ni] = frag_mfma<ET>(at_lds, bt_lds, irow, ni * 16 + l15, khalf,
-                              acc[ni]);
+      acc[ni] = frag_mfma_tswz<ET>(at_lds, bt_lds, irow, ni * 16 + l15,
+                                   khalf, acc[ni]);
     }
     __syncthreads();
   }
@@ -320,21 +393,38 @@ __global__ __launch_bounds__(kBlock) void gemm_atb_wide_kernel(
 #pragma unroll
     for (int ni = 0; ni < 8; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
+  // T14 split pipeline (see gemm_atb_kernel): 4 tile loads in flight
+  // across each MFMA phase
+  typename GemmTraits<ET>::Seg ra0[RB * 64 / kBlock / GemmTraits<ET>::kEPS];
+  typename GemmTraits<ET>::Seg ra1[RB * 64 / kBlock / GemmTraits<ET>::kEPS];
+  typename GemmTraits<ET>::Seg rb0[RB * 64 / kBlock / GemmTraits<ET>::kEPS];
+  typename GemmTraits<ET>::Seg rb1[RB * 64 / kBlock / GemmTraits<ET>::kEPS];
+  stage_loadT<ET, ALIGNED_A>(ra0, A, r_begin, r_end, Ka, i_blk, Ka);
+  stage_loadT<ET, ALIGNED_A>(ra1, A, r_begin, r_end, Ka, i_blk + BKA, Ka);
+  stage_loadT<ET, ALIGNED_B>(rb0, B, r_begin, r_end, N, n_blk, N);
+  stage_loadT<ET, ALIGNED_B>(rb1, B, r_begin, r_end, N, n_blk + BNW, N);
   for (int r0 = r_begin; r0 < r_end; r0 += RB) {
-    stage_tile_T<ET, ALIGNED_A>(at_lds, A, r0, r_end, Ka, i_blk, Ka);
-    stage_tile_T<ET, ALIGNED_A>(at_lds + BKA * PADK, A, r0, r_end, Ka,
-                                i_blk + BKA, Ka);
-    stage_tile_T<ET, ALIGNED_B>(bt_lds, B, r0, r_end, N, n_blk, N);
-    stage_tile_T<ET, ALIGNED_B>(bt_lds + BNW * PADK, B, r0, r_end, N,
-                                n_blk + BNW, N);
+    stage_writeT<ET>(at_lds, ra0);
+    stage_writeT<ET>(at_lds + BKA * PADK, ra1);
+    stage_writeT<ET>(bt_lds, rb0);
+    stage_writeT<ET>(bt_lds + BNW * PADK, rb1);
     __syncthreads();
+    if (r0 + RB < r_end) {
+      stage_loadT<ET, ALIGNED_A>(ra0, A, r0 + RB, r_end, Ka, i_blk, Ka);
+      stage_loadT<ET, ALIGNED_A>(ra1, A, r0 + RB, r_end, Ka, i_blk + BKA,
+                                 Ka);
+      stage_loadT<ET, ALIGNED_B>(rb0, B, r0 + RB, r_end, N, n_blk, N);
+      stage_loadT<ET, ALIGNED_B>(rb1, B, r0 + RB, r_end, N, n_blk + BNW,
+                                 N);
+    }
     const int irow = i_wave + l15;
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
       for (int ni = 0; ni < 8; ++ni) {
-        acc[mi][ni] = frag_mfma<ET>(at_lds + mi * BKA * PADK, bt_lds, irow,
-                                    ni * 16 + l15, khalf, acc[mi][ni]);
+        acc[mi][ni] = frag_mfma_tswz<ET>(at_lds + mi * BKA * PADK, bt_lds,
+                                         irow, ni * 16 + l15, khalf,
+                                         acc[mi][ni]);
       }
     }
     __syncthreads();
