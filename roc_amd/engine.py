@@ -97,6 +97,7 @@ class Trainer:
         if self.offload is not None:
             self.offload.prefetch()
         loss.backward()
+        F.dw_flush()  # join side-stream dW GEMMs before grad consumers
         self._allreduce_grads()
         self.optimizer.step()
         return metrics
@@ -169,6 +170,7 @@ class Trainer:
             if self.offload is not None:
                 self.offload.prefetch()
             loss.backward()
+            F.dw_flush()
             self._allreduce_grads()
             self.optimizer.step()
         else:
@@ -181,6 +183,7 @@ class Trainer:
                 if self.offload is not None:
                     self.offload.prefetch()
                 loss.backward()
+                F.dw_flush()
             with tr.span("grad_allreduce"):
                 self._allreduce_grads()
             with tr.span("adam"):
@@ -227,6 +230,7 @@ class Trainer:
         if self.offload is not None:
             self.offload.prefetch()
         self._last_loss.backward()
+        F.dw_flush()
 
     def update(self):
         self._allreduce_grads()

@@ -29,6 +29,45 @@ def has_ext() -> bool:
 
 _PHASE_COLS: Optional[int] = None
 
+# -- async dW: the weight-grad GEMMs are independent of the big SpMM^T
+# that follows them in backward, and gemm_atb ACCUMULATES (fp32
+# atomics), so when the param's .grad buffer exists (flat-grad setup)
+# the dW launches on a side stream and lands directly in .grad while
+# the aggregation backward runs. dw_flush() joins the side stream
+# before any gradient consumer (all-reduce / Adam). ROC_ASYNC_DW=0
+# reverts to the in-line synchronous dW.
+_DW_STREAM = None
+_DW_PENDING: list = []  # [(done_event, (x, dy) keep-alive refs)]
+_ASYNC_DW: Optional[bool] = None
+
+
+def _async_dw() -> bool:
+    global _ASYNC_DW
+    if _ASYNC_DW is None:
+        import os
+        _ASYNC_DW = os.environ.get("ROC_ASYNC_DW", "1") != "0"
+    return _ASYNC_DW
+
+
+def _dw_stream():
+    global _DW_STREAM
+    if _DW_STREAM is None:
+        _DW_STREAM = torch.cuda.Stream()
+    return _DW_STREAM
+
+
+def dw_flush() -> None:
+    """Make the current stream wait for every pending side-stream dW
+    GEMM. Call before anything that reads or zeroes gradients."""
+    if not _DW_PENDING:
+        return
+    from .. import streamcheck
+    cs = torch.cuda.current_stream()
+    for done, _refs in _DW_PENDING:
+        streamcheck.consumer(done, "dw-consume")
+        cs.wait_event(done)
+    _DW_PENDING.clear()
+
 
 def _phase_cols() -> int:
     """Column-phase width for strip-blocked SpMM passes (0 = off).
@@ -458,8 +497,27 @@ class _Linear(torch.autograd.Function):
                 dym = torch.empty_like(dy)
                 _C.rowscale(dym, dy, row_scale)
                 dy = dym
-            dw = torch.zeros_like(w)  # fp32 [in, out]
-            _C.gemm_atb(dw, x, dy)
+            gbuf = w.grad if _async_dw() else None
+            if (gbuf is not None and gbuf.dtype == torch.float32
+                    and gbuf.is_contiguous() and gbuf.is_cuda):
+                # accumulate straight into .grad on the side stream
+                # (gemm_atb is += via fp32 atomics); return None so
+                # autograd does not accumulate a second copy
+                from .. import streamcheck
+                s = _dw_stream()
+                ready = torch.cuda.Event()
+                ready.record()  # x/dy produced on the current stream
+                s.wait_event(ready)
+                with torch.cuda.stream(s):
+                    _C.gemm_atb(gbuf, x, dy)
+                done = torch.cuda.Event()
+                done.record(s)
+                streamcheck.producer(done, "dw-gemm")
+                _DW_PENDING.append((done, (x, dy)))
+                dw = None
+            else:
+                dw = torch.zeros_like(w)  # fp32 [in, out]
+                _C.gemm_atb(dw, x, dy)
             dx = None
             if ctx.needs_input_grad[0]:  # layer-1 inputs carry no grad
                 # dx = dy @ w^T: gemm_rr's Bt = (w^T)^T = w [in,out]
