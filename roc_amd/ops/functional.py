@@ -34,8 +34,11 @@ _PHASE_COLS: Optional[int] = None
 # atomics), so when the param's .grad buffer exists (flat-grad setup)
 # the dW launches on a side stream and lands directly in .grad while
 # the aggregation backward runs. dw_flush() joins the side stream
-# before any gradient consumer (all-reduce / Adam). ROC_ASYNC_DW=0
-# reverts to the in-line synchronous dW.
+# before any gradient consumer (all-reduce / Adam).
+# Default OFF by measurement (profiles/r29 addendum): the epoch is
+# memory-saturated end-to-end, so overlapping the memory-bound dW with
+# the memory-bound SpMM^T returns nothing (15.88 vs 15.83 ms sync) —
+# the capability stays for compute-bound model mixes (ROC_ASYNC_DW=1).
 _DW_STREAM = None
 _DW_PENDING: list = []  # [(done_event, (x, dy) keep-alive refs)]
 _ASYNC_DW: Optional[bool] = None
@@ -45,7 +48,7 @@ def _async_dw() -> bool:
     global _ASYNC_DW
     if _ASYNC_DW is None:
         import os
-        _ASYNC_DW = os.environ.get("ROC_ASYNC_DW", "1") != "0"
+        _ASYNC_DW = os.environ.get("ROC_ASYNC_DW", "0") == "1"
     return _ASYNC_DW
 
 
