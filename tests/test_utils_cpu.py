@@ -250,3 +250,29 @@ def test_predict_mode(tmp_path):
     assert "[predict] wrote" in r.stdout
     p = np.load(out)
     assert p.dtype == np.int64 and p.min() >= 0 and p.max() < 7
+
+
+def test_save_best_and_safetensors_export(tmp_path):
+    """--save-best keeps the best-val checkpoint (with the pinned val
+    metric in extra); --export-safetensors writes loadable weights."""
+    import numpy as np
+    from safetensors.torch import load_file
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    best = str(tmp_path / "best.pt")
+    st = str(tmp_path / "w.safetensors")
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "train.py"),
+         "--dataset", "cora-synthetic", "--scale", "0.2",
+         "--learnable-labels", "--epochs", "12", "--eval-every", "3",
+         "--save-best", best, "--export-safetensors", st],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "[export] wrote" in r.stdout
+    state = torch.load(best, weights_only=False)
+    assert state["extra"]["best_val_acc"] >= 0
+    assert state["extra"]["best_epoch"] % 3 == 0
+    weights = load_file(st)
+    assert weights and all(v.dtype == torch.float32
+                           for v in weights.values())
+    ck_names = set(state["model"])
+    assert set(weights) == ck_names

@@ -65,6 +65,12 @@ def parse_args():
     ap.add_argument("--resume", default=None)
     ap.add_argument("--trace", default=None,
                     help="write a chrome trace JSON here at the end")
+    ap.add_argument("--save-best", default=None,
+                    help="track the best val_acc at each eval and keep a "
+                         "checkpoint of it here")
+    ap.add_argument("--export-safetensors", default=None,
+                    help="after training, write the model weights (fp32 "
+                         "masters) as a .safetensors file")
     ap.add_argument("--predict", default=None,
                     help="inference mode: skip training, run one forward "
                          "in infer mode (dropout=identity) and write "
@@ -298,6 +304,7 @@ def main():
         return
 
     t_start = time.perf_counter()
+    best_val = -1.0
     while trainer.epoch < args.epochs:
         metrics = trainer.train_epoch()
         ep = trainer.epoch
@@ -334,6 +341,14 @@ def main():
                 print(f"[rebalance] epoch {ep}: bounds -> {nb}", flush=True)
         if args.eval_every and ep % args.eval_every == 0:
             md = trainer.evaluate()
+            if args.save_best and md["val_acc"] > best_val:
+                best_val = md["val_acc"]
+                save_checkpoint(args.save_best, trainer,
+                                extra={"best_val_acc": best_val,
+                                       "best_epoch": ep})
+                if rank == 0 and args.verbose:
+                    print(f"[best] epoch {ep}: val_acc {best_val:.4f} "
+                          f"-> {args.save_best}", flush=True)
             if rank == 0:
                 dt = time.perf_counter() - t_start
                 print(f"epoch {ep:5d}  loss {md['ce_loss']:.4f}  "
@@ -351,6 +366,12 @@ def main():
               flush=True)
     if args.checkpoint:
         save_checkpoint(args.checkpoint, trainer)
+    if args.export_safetensors and rank == 0:
+        from safetensors.torch import save_file
+        save_file({k: v.detach().float().cpu().contiguous()
+                   for k, v in trainer.model.state_dict().items()},
+                  args.export_safetensors)
+        print(f"[export] wrote {args.export_safetensors}", flush=True)
     if args.trace and trainer.tracer is not None:
         trainer.tracer.dump_chrome(args.trace, rank)
         if rank == 0:
