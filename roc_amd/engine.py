@@ -45,6 +45,7 @@ class Trainer:
         self.optimizer = optimizer
         optimizer.setup_flat_grads()  # single-buffer grads (one all-reduce)
         self.num_classes = num_classes  # true classes if logits are padded
+        self.clip_norm = 0.0  # >0: global grad-norm clip (branchless)
         self.epoch = 0
         F.set_dropout_seed(seed + shard.rank * 7919)
         self.tracer = None  # set via enable_tracing()
@@ -137,9 +138,17 @@ class Trainer:
 
     # -- gradient all-reduce (flat buffer; weights are small) ---------------
     def _allreduce_grads(self):
-        if self.shard.world_size == 1 or not dist.is_initialized():
-            return
-        dist.all_reduce(self.optimizer._flat_grad, group=self.group)
+        if self.shard.world_size > 1 and dist.is_initialized():
+            dist.all_reduce(self.optimizer._flat_grad, group=self.group)
+        if self.clip_norm > 0:
+            # branchless global-norm clip on the flat buffer: scale by
+            # clip/max(norm, clip) — 1.0 when under the bound. No host
+            # readback or data-dependent branch, so it works inside
+            # hipGraph capture (post-all-reduce: every rank scales the
+            # same summed gradient identically).
+            g = self.optimizer._flat_grad
+            norm = torch.linalg.vector_norm(g)
+            g.mul_(self.clip_norm / torch.clamp(norm, min=self.clip_norm))
 
     offload = None  # set via enable_offload()
 

@@ -167,3 +167,23 @@ def test_adaptive_agg_order_equal(name, monkeypatch):
     monkeypatch.setattr(gcn_mod, "_ADAPTIVE", None)
     assert torch.allclose(outs[0], outs[1], atol=1e-4, rtol=1e-4), \
         (outs[0] - outs[1]).abs().max()
+
+
+def test_clip_norm():
+    """Branchless global grad clip: with a tiny bound the step must
+    shrink; with a huge bound training is bit-identical to no clip."""
+    tr_a = make_trainer(lr=0.1)
+    tr_a.train_epoch()
+    w_ref = [p.detach().clone() for p in tr_a.model.parameters()]
+
+    tr_b = make_trainer(lr=0.1)
+    tr_b.clip_norm = 1e9  # never binds
+    tr_b.train_epoch()
+    for a, b in zip(w_ref, tr_b.model.parameters()):
+        assert torch.equal(a, b.detach())
+
+    tr_c = make_trainer(lr=0.1)
+    tr_c.clip_norm = 1e-3  # always binds
+    tr_c.train_epoch()
+    gnorm = tr_c.optimizer._flat_grad.norm().item()
+    assert gnorm <= 1e-3 * 1.01, gnorm

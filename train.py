@@ -101,6 +101,9 @@ def parse_args():
     ap.add_argument("--rebalance-every", type=int, default=0,
                     help="cost-model repartition every N epochs (measured "
                          "per-rank throughput; multi-rank only)")
+    ap.add_argument("--clip-norm", type=float, default=0.0,
+                    help="global gradient-norm clip (0 = off); applied to "
+                         "the all-reduced flat gradient on every rank")
     ap.add_argument("--auto-recover", action="store_true",
                     help="on divergence (non-finite loss): restore the "
                          "last checkpoint, halve the LR, continue")
@@ -237,6 +240,7 @@ def main():
                       compute_dtype=dtype, grad_scale=gs, seed=args.seed,
                       num_classes=num_classes if dims[-1] != num_classes
                       else None, local_slices=pre_shard is not None)
+    trainer.clip_norm = args.clip_norm
     if args.trace:
         trainer.enable_tracing()
     if args.offload:
