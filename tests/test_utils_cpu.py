@@ -276,3 +276,21 @@ def test_save_best_and_safetensors_export(tmp_path):
                            for v in weights.values())
     ck_names = set(state["model"])
     assert set(weights) == ck_names
+
+
+def test_log_csv(tmp_path):
+    """--log-csv appends a well-formed metrics row per eval."""
+    import csv
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    log = str(tmp_path / "m.csv")
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "train.py"),
+         "--dataset", "cora-synthetic", "--scale", "0.2",
+         "--epochs", "9", "--eval-every", "3", "--log-csv", log],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    with open(log) as f:
+        rows = list(csv.DictReader(f))
+    assert [int(x["epoch"]) for x in rows] == [3, 6, 9]
+    assert all(float(x["ce_loss"]) > 0 for x in rows)
+    assert all(float(x["ms_per_epoch"]) > 0 for x in rows)

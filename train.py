@@ -101,6 +101,9 @@ def parse_args():
     ap.add_argument("--rebalance-every", type=int, default=0,
                     help="cost-model repartition every N epochs (measured "
                          "per-rank throughput; multi-rank only)")
+    ap.add_argument("--log-csv", default=None,
+                    help="append one CSV row per eval (epoch, losses, "
+                         "accuracies, ms/epoch); rank 0 only")
     ap.add_argument("--clip-norm", type=float, default=0.0,
                     help="global gradient-norm clip (0 = off); applied to "
                          "the all-reduced flat gradient on every rank")
@@ -359,6 +362,18 @@ def main():
                       f"train {md['train_acc']:.4f}  val {md['val_acc']:.4f}"
                       f"  test {md['test_acc']:.4f}  "
                       f"[{dt / ep * 1e3:.1f} ms/epoch]", flush=True)
+                if args.log_csv:
+                    new = not os.path.exists(args.log_csv)
+                    with open(args.log_csv, "a") as fcsv:
+                        if new:
+                            fcsv.write("epoch,ce_loss,roc_loss,train_acc,"
+                                       "val_acc,test_acc,ms_per_epoch\n")
+                        fcsv.write(f"{ep},{md['ce_loss']:.6f},"
+                                   f"{md.get('roc_loss', 0.0):.6f},"
+                                   f"{md['train_acc']:.6f},"
+                                   f"{md['val_acc']:.6f},"
+                                   f"{md['test_acc']:.6f},"
+                                   f"{dt / ep * 1e3:.3f}\n")
         if (args.checkpoint and args.checkpoint_every
                 and ep % args.checkpoint_every == 0):
             save_checkpoint(args.checkpoint, trainer)
