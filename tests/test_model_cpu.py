@@ -20,7 +20,7 @@ def make_trainer(model_name="gcn", scale=0.05, hidden=16, lr=0.01,
                    grad_scale=grad_scale)
 
 
-@pytest.mark.parametrize("name", ["gcn", "sage", "gin"])
+@pytest.mark.parametrize("name", ["gcn", "sage", "gin", "sgc", "appnp"])
 def test_model_trains(name):
     tr = make_trainer(name)
     m0 = tr.evaluate()

@@ -23,7 +23,7 @@ def make_trainer(name="gcn", scale=0.3, hidden=64, dtype=torch.bfloat16,
                    compute_dtype=dtype, grad_scale=gs)
 
 
-@pytest.mark.parametrize("name", ["gcn", "sage", "gin"])
+@pytest.mark.parametrize("name", ["gcn", "sage", "gin", "sgc", "appnp"])
 def test_model_trains_gpu(name):
     tr = make_trainer(name)
     m0 = tr.evaluate()

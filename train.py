@@ -45,7 +45,7 @@ def parse_args():
                          "e.g. 602-256-41")
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--num-layers", type=int, default=2)
-    ap.add_argument("--model", default="gcn", choices=["gcn", "sage", "gin", "gat"])
+    ap.add_argument("--model", default="gcn", choices=["gcn", "sage", "gin", "gat", "sgc", "appnp"])
     ap.add_argument("--epochs", "-e", type=int, default=100)
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--weight-decay", "--wd", type=float, default=1e-4)
