@@ -348,7 +348,7 @@ def _widen_worker(rank, port, mode, model_name, q):
 
 @pytest.mark.parametrize("mode,model_name,port", [
     ("halo", "gcn", 29531), ("allgather", "gcn", 29533),
-    ("halo", "sage", 29535),
+    ("halo", "sage", 29535), ("halo", "appnp", 29537),
 ])
 def test_widening_layer_sharded_matches_single(mode, model_name, port):
     ctx = mp.get_context("spawn")
