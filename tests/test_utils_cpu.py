@@ -3,6 +3,8 @@ import os
 import subprocess
 import sys
 
+import pytest
+
 import torch
 
 from roc_amd import (synthetic_dataset, build_shard, build_model,
@@ -294,3 +296,23 @@ def test_log_csv(tmp_path):
     assert [int(x["epoch"]) for x in rows] == [3, 6, 9]
     assert all(float(x["ce_loss"]) > 0 for x in rows)
     assert all(float(x["ms_per_epoch"]) > 0 for x in rows)
+
+
+@pytest.mark.parametrize("model", ["sgc", "appnp"])
+def test_new_family_checkpoint_predict(tmp_path, model):
+    """SGC/APPNP round-trip through checkpoint -> resume -> --predict."""
+    import numpy as np
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ck = str(tmp_path / "ck.pt")
+    out = str(tmp_path / "p.npy")
+    base = [sys.executable, os.path.join(repo, "train.py"),
+            "--dataset", "cora-synthetic", "--scale", "0.2",
+            "--model", model, "--eval-every", "0"]
+    r = subprocess.run(base + ["--epochs", "3", "--checkpoint", ck],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    r = subprocess.run(base + ["--resume", ck, "--predict", out],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    p = np.load(out)
+    assert p.dtype == np.int64 and p.min() >= 0 and p.max() < 7
