@@ -201,3 +201,32 @@ def test_async_dw_matches_sync():
         # fp32 atomic accumulation order differs run-to-run either way;
         # async must stay inside that noise floor
         assert torch.allclose(a, b, atol=2e-3), (a - b).abs().max()
+
+
+@pytest.mark.parametrize("name", ["sgc", "appnp"])
+def test_new_family_gpu_matches_cpu(name):
+    """SGC / APPNP: one epoch GPU bf16 vs CPU fp32 (no dropout) — the
+    precomputed propagation (SGC) and the K-hop differentiable
+    propagation (APPNP) must agree across backends."""
+    g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.2, seed=3)
+    pad = (-feats.shape[1]) % 8
+    if pad:
+        feats = torch.nn.functional.pad(feats, (0, pad))
+    shard = build_shard(g, 0, 1)
+    dims = [feats.shape[1], 32, c]
+
+    results = {}
+    for dev, dtype in (("cpu", torch.float32), ("cuda:0", torch.bfloat16)):
+        model = build_model(name, dims, dropout=0.0, seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
+        tr = Trainer(model, shard, feats, labels, mask, opt, device=dev,
+                     compute_dtype=dtype)
+        tr.train_epoch()
+        results[dev] = (tr.evaluate(),
+                        next(model.parameters()).detach().cpu().clone())
+    m_cpu, w_cpu = results["cpu"]
+    m_gpu, w_gpu = results["cuda:0"]
+    assert m_gpu["ce_loss"] == pytest.approx(m_cpu["ce_loss"], rel=0.07)
+    assert m_gpu["train_total"] == m_cpu["train_total"]
+    assert torch.allclose(w_gpu, w_cpu, atol=0.05), \
+        (w_gpu - w_cpu).abs().max()
