@@ -1,6 +1,8 @@
 """Native C++ graph machinery (roc_amd._C CPU functions) vs numpy."""
 import numpy as np
 import pytest
+import os
+
 import torch
 
 from roc_amd.graph import synthetic_graph
@@ -73,3 +75,38 @@ def test_sync_debug_mode():
     finally:
         enable_sync_debug(False)
     assert not sync_debug_enabled()
+
+
+def test_knobs_doc_covers_all_env_vars():
+    """docs/KNOBS.md must document every ROC_* environment variable the
+    code actually reads (doc-rot guard)."""
+    import re
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    read_pat = re.compile(
+        r'(?:environ(?:\.get)?\s*[\[(]\s*"|getenv\s*\(\s*")(ROC_[A-Z0-9_]+)')
+    used = set()
+    for root, _dirs, files in os.walk(os.path.join(repo, "roc_amd")):
+        for fn in files:
+            if fn.endswith((".py", ".hip", ".cpp", ".h")):
+                with open(os.path.join(root, fn), errors="ignore") as f:
+                    used |= set(read_pat.findall(f.read()))
+    for fn in ("train.py", "bench.py"):
+        with open(os.path.join(repo, fn)) as f:
+            used |= set(read_pat.findall(f.read()))
+    with open(os.path.join(repo, "docs", "KNOBS.md")) as f:
+        documented = set(re.findall(r"ROC_[A-Z0-9_]+", f.read()))
+    missing = used - documented
+    assert not missing, f"undocumented env knobs: {sorted(missing)}"
+
+
+def test_profiles_index_covers_all_reports():
+    """profiles/INDEX.md must reference every rNN report file (keeps the
+    evidence chain navigable)."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    pdir = os.path.join(repo, "profiles")
+    with open(os.path.join(pdir, "INDEX.md")) as f:
+        idx = f.read()
+    missing = [fn for fn in sorted(os.listdir(pdir))
+               if fn.startswith("r") and fn not in idx
+               and fn != "INDEX.md"]
+    assert not missing, f"unindexed profile reports: {missing}"
