@@ -18,19 +18,19 @@ from roc_amd.parallel.partition import build_shard, edge_balanced_bounds
 WS = 8
 
 
-def _worker(rank, port, mode, q):
+def _worker(rank, port, mode, q, ws=WS):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
         os.environ["ROC_COMM_MODE"] = mode
         if mode == "halo":
             os.environ["ROC_OVERLAP"] = "1"  # overlap split path at ws=8
-        dist.init_process_group("gloo", rank=rank, world_size=WS)
+        dist.init_process_group("gloo", rank=rank, world_size=ws)
         torch.manual_seed(0)
         g, feats, labels, mask, c = synthetic_dataset("cora", scale=0.2,
                                                       seed=3)
-        bounds = edge_balanced_bounds(g.rowptr, WS)
-        sh = build_shard(g, rank, WS, bounds)
+        bounds = edge_balanced_bounds(g.rowptr, ws)
+        sh = build_shard(g, rank, ws, bounds)
         model = build_model("gcn", [feats.shape[1], 16, c], dropout=0.0,
                             seed=1)
         opt = AdamOptimizer(model.parameters(), lr=0.01, weight_decay=1e-4)
@@ -87,22 +87,24 @@ def test_bench_contract_ws8_gloo():
     assert "x8" in out["config"]["parallelism"]
 
 
-@pytest.mark.parametrize("mode", ["halo", "allgather"])
-def test_ws8_matches_single_rank(mode):
+@pytest.mark.parametrize("mode,nranks,port", [
+    ("halo", 8, 29561), ("allgather", 8, 29563),
+    ("halo", 4, 29565), ("allgather", 4, 29567),  # the driver's N=4 point
+])
+def test_ws8_matches_single_rank(mode, nranks, port):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29561 if mode == "halo" else 29563
-    procs = [ctx.Process(target=_worker, args=(r, port, mode, q))
-             for r in range(WS)]
+    procs = [ctx.Process(target=_worker, args=(r, port, mode, q, nranks))
+             for r in range(nranks)]
     for p in procs:
         p.start()
-    res = sorted([q.get() for _ in range(WS)], key=lambda t: t[0])
+    res = sorted([q.get() for _ in range(nranks)], key=lambda t: t[0])
     for p in procs:
         p.join(timeout=300)
     for rank, md, w, err in res:
         assert err is None, f"rank {rank}: {err}"
     ws = [torch.from_numpy(r[2]) for r in res]
-    for k in range(1, WS):  # replicated weights identical everywhere
+    for k in range(1, nranks):  # replicated weights identical everywhere
         assert torch.allclose(ws[0], ws[k], atol=1e-6), k
     md1, w1 = _single_rank_baseline()
     assert torch.allclose(w1, ws[0], atol=1e-4), (w1 - ws[0]).abs().max()
