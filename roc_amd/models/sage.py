@@ -63,3 +63,25 @@ class GraphSAGE(torch.nn.Module):
         if i < len(self.w_self) - 1:
             h = F.relu(h)
         return h
+
+    def forward_blocks(self, x, blocks):
+        """Mini-batch forward over sampled MFG blocks (roc_amd.sampling):
+        layer i aggregates over blocks[i]; x holds the input features
+        for blocks[0].src_ids rows. Returns [batch, dims[-1]]."""
+        from .gcn import _adaptive_agg
+        assert len(blocks) == len(self.w_self), (len(blocks), self.dims)
+        for i in range(len(self.w_self)):
+            blk = blocks[i]
+            h = F.dropout(x, self.p, self.training, call_id=i)
+            h_self = F.linear(h[:blk.n_dst].contiguous(), self.w_self[i])
+            wn = self.w_neigh[i]
+            if wn.shape[0] < wn.shape[1] and _adaptive_agg():
+                hn = F.scatter_gather(h, blk, dst_scale=blk.inv_deg)
+                hn = F.linear(hn, wn)
+            else:
+                hn = F.linear(h, wn)
+                hn = F.scatter_gather(hn, blk, dst_scale=blk.inv_deg)
+            x = F.add(h_self, hn)
+            if i < len(self.w_self) - 1:
+                x = F.relu(x)
+        return x

@@ -1,0 +1,205 @@
+"""Mini-batch neighbor-sampled training (GraphSAGE-style MFG blocks).
+
+The reference (and this framework's headline path) trains FULL-graph —
+that is ROC's thesis. This module adds the complementary production
+mode: per-batch K-hop neighbor sampling, so enormous graphs can train
+with bounded memory per step and the serving tier can embed fresh
+nodes. It is deliberately layered ON TOP of the existing compute path:
+a sampled batch materializes one `Block` (bipartite CSR: dst rows x
+src cols) per hop, and each Block quacks like a GraphShard, so the
+SAME `scatter_gather` autograd op and the SAME CDNA4 SpMM kernels run
+over it — no new kernels, no parallel code path to keep correct.
+
+Sampling itself is a numpy routine (without-replacement per row, seeded
+rng). It runs on the host while the previous batch computes; a native
+sampler is the obvious next step if profiles ever show it on the
+critical path.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, replace
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from .graph import CSRGraph, build_transpose
+
+
+@dataclass
+class Block:
+    """One hop's bipartite aggregation graph: out rows are the DST
+    nodes (first n_dst of the src id space), columns are SRC nodes.
+    Field names/shapes match what ops.functional.scatter_gather reads
+    from a GraphShard, so Blocks drop into the existing op."""
+
+    n_local: int            # n_dst
+    n_ext: int              # n_src (dst nodes occupy src ids [0, n_dst))
+    rowptr: torch.Tensor    # int64 [n_dst + 1]
+    colidx: torch.Tensor    # int32 [n_sampled_edges], src-local ids
+    t_rowptr: torch.Tensor  # transpose (src-rows) for the backward
+    t_colidx: torch.Tensor
+    src_ids: torch.Tensor   # int64 [n_src] global ids (layer input rows)
+    inv_deg: torch.Tensor   # fp32 [n_dst], 1/sampled-indegree (mean agg)
+    # GraphShard-compat fields the op reads but blocks never use:
+    row_order = None
+    t_row_order = None
+    fwd_strips = None
+    bwd_strips = None
+
+    @property
+    def n_dst(self) -> int:
+        return self.n_local
+
+    @property
+    def n_src(self) -> int:
+        return self.n_ext
+
+    def to(self, device) -> "Block":
+        return replace(
+            self,
+            rowptr=self.rowptr.to(device), colidx=self.colidx.to(device),
+            t_rowptr=self.t_rowptr.to(device),
+            t_colidx=self.t_colidx.to(device),
+            src_ids=self.src_ids.to(device), inv_deg=self.inv_deg.to(device))
+
+
+def sample_blocks(g: CSRGraph, targets: np.ndarray,
+                  fanouts: Sequence[int],
+                  rng: Optional[np.random.Generator] = None) -> List[Block]:
+    """Sample K-hop in-neighborhoods for `targets` (global node ids).
+
+    fanouts are OUTER-TO-INNER per model layer (fanouts[0] serves layer
+    0, the hop closest to the input features — same convention as the
+    dims list). Returns blocks[0..K-1] in LAYER order: the layer-i
+    aggregation runs over blocks[i]; blocks[-1].n_dst == len(targets)
+    and blocks[0].src_ids names the input-feature rows to load.
+    Each dst row keeps min(fanout, indegree) distinct in-neighbors
+    (without replacement) and always includes itself in the src set.
+    """
+    if rng is None:
+        rng = np.random.default_rng(0)
+    rp = g.rowptr.numpy()
+    ci = g.colidx.numpy()
+    blocks: List[Block] = []
+    dst = np.asarray(targets, dtype=np.int64)
+    # build from the OUTPUT side inward, then reverse into layer order
+    for fanout in reversed(list(fanouts)):
+        n_dst = dst.shape[0]
+        # dst nodes take src-local ids [0, n_dst); sampled extras follow
+        local = {int(v): i for i, v in enumerate(dst)}
+        rows = []
+        cols = []
+        extras: List[int] = []
+        for i, v in enumerate(dst):
+            lo, hi = int(rp[v]), int(rp[v + 1])
+            deg = hi - lo
+            if deg <= 0:
+                rows.append(0)
+                continue
+            if deg <= fanout:
+                picked = ci[lo:hi]
+            else:
+                picked = ci[lo + rng.choice(deg, size=fanout,
+                                            replace=False)]
+            rows.append(len(picked))
+            for u in picked:
+                u = int(u)
+                j = local.get(u)
+                if j is None:
+                    j = n_dst + len(extras)
+                    local[u] = j
+                    extras.append(u)
+                cols.append(j)
+        rowptr = torch.zeros(n_dst + 1, dtype=torch.int64)
+        torch.cumsum(torch.tensor(rows, dtype=torch.int64), 0,
+                     out=rowptr[1:])
+        colidx = torch.tensor(cols, dtype=torch.int32)
+        src_ids = torch.from_numpy(
+            np.concatenate([dst, np.asarray(extras, dtype=np.int64)]))
+        n_src = int(src_ids.numel())
+        t_rowptr, t_colidx = build_transpose(n_src, rowptr, colidx)
+        deg_t = (rowptr[1:] - rowptr[:-1]).clamp(min=1).to(torch.float32)
+        blocks.append(Block(n_local=n_dst, n_ext=n_src, rowptr=rowptr,
+                            colidx=colidx, t_rowptr=t_rowptr,
+                            t_colidx=t_colidx, src_ids=src_ids,
+                            inv_deg=1.0 / deg_t))
+        dst = src_ids.numpy()  # next (inner) hop samples for all srcs
+    blocks.reverse()
+    return blocks
+
+
+class MiniBatchTrainer:
+    """Sampled-minibatch training loop (GraphSAGE-style). Features,
+    labels and the graph stay on the HOST; only each batch's sampled
+    rows and blocks move to the device — bounded device memory per
+    step regardless of graph size. Evaluation runs the model's
+    full-graph forward (the standard offline protocol)."""
+
+    def __init__(self, model, g: CSRGraph, feats, labels, mask, optimizer,
+                 fanouts: Sequence[int], batch_size: int = 1024,
+                 device="cpu", compute_dtype: torch.dtype = torch.float32,
+                 seed: int = 1, num_classes: Optional[int] = None):
+        from .ops import functional as F
+        assert hasattr(model, "forward_blocks"), \
+            "mini-batch training needs a model with forward_blocks (sage)"
+        self.model = model.to(device)
+        self.g = g
+        self.feats = feats          # host-resident [N, D]
+        self.labels = labels        # host-resident [N]
+        self.mask = mask
+        self.optimizer = optimizer
+        optimizer.setup_flat_grads()
+        self.fanouts = list(fanouts)
+        self.batch_size = int(batch_size)
+        self.device = torch.device(device)
+        self.dtype = compute_dtype
+        self.train_ids = np.nonzero(mask.numpy() == 1)[0]
+        assert self.train_ids.size, "no Train-masked nodes"
+        self.rng = np.random.default_rng(seed)
+        self.epoch = 0
+        self.num_classes = num_classes
+        F.set_dropout_seed(seed)
+
+    def train_epoch(self) -> float:
+        from .ops import functional as F
+        self.model.train()
+        order = self.rng.permutation(self.train_ids)
+        total, nb = 0.0, 0
+        ones = None
+        for s in range(0, order.size, self.batch_size):
+            batch = order[s:s + self.batch_size]
+            blocks = [b.to(self.device)
+                      for b in sample_blocks(self.g, batch, self.fanouts,
+                                             self.rng)]
+            bt = torch.from_numpy(batch)
+            x = self.feats[blocks[0].src_ids.cpu()].to(
+                device=self.device, dtype=self.dtype)
+            yb = self.labels[bt].to(self.device)
+            if ones is None or ones.numel() != yb.numel():
+                ones = torch.ones(yb.numel(), dtype=torch.int32,
+                                  device=self.device)
+            self.optimizer.zero_grad()
+            logits = self.model.forward_blocks(x, blocks)
+            loss, metrics = F.softmax_cross_entropy(
+                logits, yb, ones, 1.0 / yb.numel(), self.num_classes)
+            loss.backward()
+            F.dw_flush()
+            self.optimizer.step()
+            total += float(metrics[1]) / yb.numel()
+            nb += 1
+        self.epoch += 1
+        return total / max(nb, 1)
+
+    @torch.no_grad()
+    def evaluate(self, shard) -> dict:
+        """Full-graph eval (standard protocol: sampled train, exact
+        inference) — needs a GraphShard of the same graph."""
+        from .ops import functional as F
+        self.model.eval()
+        x = self.feats.to(device=self.device, dtype=self.dtype)
+        logits = self.model(x, shard)
+        _, metrics = F.softmax_cross_entropy(
+            logits, self.labels.to(self.device),
+            self.mask.to(self.device, torch.int32), 1.0, self.num_classes)
+        return F.decode_metrics(metrics)

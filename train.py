@@ -101,6 +101,12 @@ def parse_args():
     ap.add_argument("--rebalance-every", type=int, default=0,
                     help="cost-model repartition every N epochs (measured "
                          "per-rank throughput; multi-rank only)")
+    ap.add_argument("--sample", default=None,
+                    help="mini-batch neighbor sampling: per-layer fanouts "
+                         "e.g. '25,10' (layer order, input-side first); "
+                         "GraphSAGE only, world_size=1")
+    ap.add_argument("--batch-size", type=int, default=1024,
+                    help="target nodes per sampled mini-batch")
     ap.add_argument("--log-csv", default=None,
                     help="append one CSV row per eval (epoch, losses, "
                          "accuracies, ms/epoch); rank 0 only")
@@ -239,6 +245,40 @@ def main():
         from roc_amd.parallel.comm import allreduce_scalar_int
         n_train = allreduce_scalar_int(n_train)
     gs = 1.0 if args.loss == "sum" else 1.0 / max(n_train, 1)
+    if args.sample:
+        # sampled mini-batch tier (roc_amd/sampling.py): host-resident
+        # graph+features, bounded device memory per step
+        from roc_amd.sampling import MiniBatchTrainer
+        assert world == 1, "--sample is a single-process mode"
+        assert args.model == "sage", "--sample needs --model sage"
+        assert g is not None, "--sample needs a full-graph dataset"
+        fanouts = [int(f) for f in args.sample.split(",")]
+        assert len(fanouts) == len(dims) - 1, \
+            f"need one fanout per layer ({len(dims) - 1})"
+        mb = MiniBatchTrainer(model, g, feats, labels, mask, opt,
+                              fanouts=fanouts, batch_size=args.batch_size,
+                              device=device, compute_dtype=dtype,
+                              seed=args.seed, num_classes=num_classes
+                              if dims[-1] != num_classes else None)
+        import time as _time
+        t0 = _time.perf_counter()
+        for ep in range(1, args.epochs + 1):
+            loss = mb.train_epoch()
+            if args.eval_every and ep % args.eval_every == 0:
+                md = mb.evaluate(shard)
+                dt = _time.perf_counter() - t0
+                print(f"epoch {ep:5d}  batch-loss {loss:.4f}  "
+                      f"train {md['train_acc']:.4f}  "
+                      f"val {md['val_acc']:.4f}  "
+                      f"[{dt / ep * 1e3:.1f} ms/epoch]", flush=True)
+        if args.checkpoint:
+            trainer = Trainer(model, shard, feats, labels, mask, opt,
+                              device=device, compute_dtype=dtype,
+                              seed=args.seed)
+            trainer.epoch = args.epochs
+            save_checkpoint(args.checkpoint, trainer)
+        return
+
     trainer = Trainer(model, shard, feats, labels, mask, opt, device=device,
                       compute_dtype=dtype, grad_scale=gs, seed=args.seed,
                       num_classes=num_classes if dims[-1] != num_classes
