@@ -232,6 +232,9 @@ def test_new_family_gpu_matches_cpu(name):
         (w_gpu - w_cpu).abs().max()
 
 
+DEV = "cuda:0"
+
+
 def test_minibatch_sampled_gpu():
     """Sampled MFG blocks run the same HIP kernels: one sampled epoch
     on GPU bf16 trains, and the full-fanout block forward matches the
