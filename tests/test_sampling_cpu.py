@@ -94,3 +94,53 @@ def test_sample_cli(tmp_path):
     assert r.returncode == 0, r.stderr[-1000:]
     assert "epoch     4" in r.stdout, r.stdout[-500:]
     assert (tmp_path / "ck.pt").exists()
+
+
+def test_native_sampler_matches_python_at_full_fanout():
+    """At fanout >= max degree both implementations keep every edge in
+    the same order -> bit-identical blocks; under sampling the native
+    hop (graph_cpu.cpp) is deterministic in its seed."""
+    from roc_amd import _C
+    from roc_amd.sampling import _sample_hop
+    g, *_ = _graph()
+    targets = np.arange(0, g.num_nodes, 5, dtype=np.int64)
+    max_deg = int((g.rowptr[1:] - g.rowptr[:-1]).max())
+    rp_n, ci_n, src_n = _C.sample_hop(g.rowptr, g.colidx,
+                                      torch.from_numpy(targets),
+                                      max_deg, 123)
+    rng = np.random.default_rng(0)
+    # force the python path by exceeding the native fanout clamp? no —
+    # call the fallback body via a fanout the clamp allows but compare
+    # to an independently-built python hop:
+    import roc_amd.sampling as S
+    rp_p, ci_p, src_p = None, None, None
+    # python reference inline (full fanout keeps all edges in order)
+    rpn = g.rowptr.numpy(); cin = g.colidx.numpy()
+    local = {int(v): i for i, v in enumerate(targets)}
+    rows, cols, extras = [], [], []
+    for v in targets:
+        seg = cin[rpn[v]:rpn[v + 1]]
+        rows.append(len(seg))
+        for u in seg:
+            u = int(u)
+            j = local.get(u)
+            if j is None:
+                j = len(targets) + len(extras)
+                local[u] = j
+                extras.append(u)
+            cols.append(j)
+    rp_p = torch.zeros(len(targets) + 1, dtype=torch.int64)
+    torch.cumsum(torch.tensor(rows, dtype=torch.int64), 0, out=rp_p[1:])
+    ci_p = torch.tensor(cols, dtype=torch.int32)
+    src_p = torch.from_numpy(np.concatenate(
+        [targets, np.asarray(extras, dtype=np.int64)]))
+    assert torch.equal(rp_n, rp_p)
+    assert torch.equal(ci_n, ci_p)
+    assert torch.equal(src_n, src_p)
+    # determinism under real sampling: same seed -> same result
+    a = _C.sample_hop(g.rowptr, g.colidx, torch.from_numpy(targets), 4, 7)
+    b = _C.sample_hop(g.rowptr, g.colidx, torch.from_numpy(targets), 4, 7)
+    for x, y in zip(a, b):
+        assert torch.equal(x, y)
+    c = _C.sample_hop(g.rowptr, g.colidx, torch.from_numpy(targets), 4, 8)
+    assert not all(torch.equal(x, y) for x, y in zip(a, c))
