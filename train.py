@@ -60,6 +60,11 @@ def parse_args():
     ap.add_argument("--residual", action="store_true")
     ap.add_argument("--heads", type=int, default=4,
                     help="GAT attention heads (concat; last layer 1)")
+    ap.add_argument("--k-hops", type=int, default=None,
+                    help="SGC propagation depth (default 2) / APPNP "
+                         "propagation steps (default 10)")
+    ap.add_argument("--alpha", type=float, default=0.1,
+                    help="APPNP teleport probability")
     ap.add_argument("--checkpoint", default=None)
     ap.add_argument("--checkpoint-every", type=int, default=0)
     ap.add_argument("--resume", default=None)
@@ -224,6 +229,12 @@ def main():
     mkw = {"residual": args.residual} if args.model == "gcn" else {}
     if args.model == "gat":
         mkw = {"heads": args.heads}
+    elif args.model == "sgc" and args.k_hops:
+        mkw = {"k": args.k_hops}
+    elif args.model == "appnp":
+        mkw = {"alpha": args.alpha}
+        if args.k_hops:
+            mkw["k"] = args.k_hops
     model = build_model(args.model, dims, dropout=args.dropout,
                         seed=args.seed, **mkw)
     if args.recompute:
