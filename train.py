@@ -263,6 +263,11 @@ def main():
         assert world == 1, "--sample is a single-process mode"
         assert args.model == "sage", "--sample needs --model sage"
         assert g is not None, "--sample needs a full-graph dataset"
+        assert not (args.predict or args.save_best or args.log_csv or
+                    args.resume), (
+            "--predict/--save-best/--log-csv/--resume are full-graph-loop "
+            "features: train with --sample --checkpoint, then run them in "
+            "a separate invocation without --sample"
         fanouts = [int(f) for f in args.sample.split(",")]
         assert len(fanouts) == len(dims) - 1, \
             f"need one fanout per layer ({len(dims) - 1})"
