@@ -267,7 +267,7 @@ def main():
                     args.resume), (
             "--predict/--save-best/--log-csv/--resume are full-graph-loop "
             "features: train with --sample --checkpoint, then run them in "
-            "a separate invocation without --sample"
+            "a separate invocation without --sample")
         fanouts = [int(f) for f in args.sample.split(",")]
         assert len(fanouts) == len(dims) - 1, \
             f"need one fanout per layer ({len(dims) - 1})"
