@@ -316,3 +316,18 @@ def test_new_family_checkpoint_predict(tmp_path, model):
     assert r.returncode == 0, r.stderr[-800:]
     p = np.load(out)
     assert p.dtype == np.int64 and p.min() >= 0 and p.max() < 7
+
+
+@pytest.mark.parametrize("extra", [
+    ["--model", "sgc", "--k-hops", "3"],
+    ["--model", "appnp", "--k-hops", "4", "--alpha", "0.2"],
+])
+def test_model_hyperparam_flags(extra):
+    """--k-hops / --alpha reach the model constructors."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "train.py"),
+         "--dataset", "cora-synthetic", "--scale", "0.2",
+         "--epochs", "2", "--eval-every", "0", "-v"] + extra,
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
