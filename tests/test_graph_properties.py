@@ -120,3 +120,37 @@ def test_halo_plan_invariants(g, world):
                     want = sh_s.halo_ids[ho:ho + rc].numpy()
                 ho += rc
             assert np.array_equal(np.sort(sent), np.sort(want)), (r, s)
+
+
+@settings(max_examples=40, deadline=None, derandomize=True)
+@given(graphs(), st.integers(min_value=1, max_value=12),
+       st.integers(min_value=0, max_value=2**31 - 1))
+def test_native_sampler_invariants(g, fanout, seed):
+    """sample_hop contracts for ANY graph/fanout/seed: degrees bounded
+    by min(fanout, true degree), every sampled edge exists in the
+    graph, targets form the src-id prefix, ids in range."""
+    from roc_amd import _C
+    rng = np.random.default_rng(seed)
+    k = min(g.num_nodes, 1 + seed % 16)
+    targets = torch.from_numpy(
+        rng.choice(g.num_nodes, size=k, replace=False).astype(np.int64))
+    rp_s, ci_s, src = _C.sample_hop(g.rowptr, g.colidx, targets,
+                                    fanout, seed)
+    assert rp_s.numel() == k + 1 and int(rp_s[0]) == 0
+    assert torch.equal(src[:k], targets)
+    assert src.unique().numel() == src.numel()  # local ids are distinct
+    rp = g.rowptr.numpy()
+    ci = g.colidx.numpy()
+    for i in range(k):
+        v = int(targets[i])
+        deg_s = int(rp_s[i + 1] - rp_s[i])
+        deg_true = int(rp[v + 1] - rp[v])
+        assert deg_s == min(fanout, deg_true)
+        neigh = ci[rp[v]:rp[v + 1]].tolist()
+        picked = [int(src[int(ci_s[e])])
+                  for e in range(int(rp_s[i]), int(rp_s[i + 1]))]
+        for u in picked:
+            assert u in neigh
+        if deg_true > fanout:  # sampled WITHOUT replacement
+            assert len(set(picked)) == len(picked) or \
+                len(set(neigh)) < len(neigh)  # unless graph has dup edges
