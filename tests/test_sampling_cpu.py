@@ -144,3 +144,19 @@ def test_native_sampler_matches_python_at_full_fanout():
         assert torch.equal(x, y)
     c = _C.sample_hop(g.rowptr, g.colidx, torch.from_numpy(targets), 4, 8)
     assert not all(torch.equal(x, y) for x, y in zip(a, c))
+
+
+def test_minibatch_determinism():
+    """Same seed -> identical loss trajectory (native sampler streams
+    are deterministic; dropout/init are seeded)."""
+    def run():
+        g, feats, labels, mask, c = _graph()
+        torch.manual_seed(0)
+        model = build_model("sage", [feats.shape[1], 16, c],
+                            dropout=0.3, seed=1)
+        opt = AdamOptimizer(model.parameters(), lr=0.02)
+        tr = MiniBatchTrainer(model, g, feats, labels, mask, opt,
+                              fanouts=[6, 6], batch_size=64, seed=9)
+        return [tr.train_epoch() for _ in range(3)]
+    a, b = run(), run()
+    assert a == b, (a, b)
